@@ -1,0 +1,142 @@
+"""Flagship benchmark: GCBF training throughput on DubinsCar n=16.
+
+Measures the BASELINE.json headline metric — env-steps/sec (whole node) on
+the paper config (DubinsCar, 16 agents, GCBF, batch_size 512) with synthetic
+random-init agents/goals (the environment itself is the synthetic data
+source; there are no datasets).  A "step" is one full training step: env
+rollout step + buffer append + the amortized `update` (10 inner iters of
+4 GNN forwards + backward + Adam on a ~300-graph batch) every
+``batch_size`` steps, exactly as `train.py` runs it.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Weak scaling: each rank runs its own env replica; gradients all-reduce over
+RCCL/xGMI each inner iteration; whole-node value = N · K / max_rank_time.
+"""
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+from gcbf_amd.algo import make_algo
+from gcbf_amd.env import make_env
+from gcbf_amd.parallel import (GradSynchronizer, broadcast_modules,
+                               cleanup_distributed, init_distributed)
+from gcbf_amd.trainer.utils import read_params, set_seed
+
+
+def _barrier_sync(device):
+    if torch.distributed.is_initialized():
+        torch.distributed.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize(device)
+
+
+def run(args):
+    rank, world_size, local_rank = init_distributed()
+    if world_size != args.gpus and rank == 0:
+        print(f"# note: WORLD_SIZE={world_size} != --gpus {args.gpus}; "
+              f"using WORLD_SIZE", flush=True)
+        args.gpus = world_size
+    set_seed(args.seed + rank)
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_cuda \
+        else torch.device("cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    dtype = args.dtype or ("bf16" if use_cuda else "fp32")
+
+    params_env = make_env(args.env, args.num_agents, device).default_params
+    params_env["num_obs"] = args.obs
+    env = make_env(args.env, args.num_agents, device, params=params_env)
+    env.train()
+    hyper = read_params(args.env, "gcbf")
+    algo = make_algo("gcbf", env, args.num_agents, env.node_dim,
+                     env.edge_dim, env.action_dim, device,
+                     batch_size=args.batch_size, hyperparams=hyper)
+    if world_size > 1:
+        broadcast_modules([algo.cbf, algo.actor])
+        algo.grad_sync = GradSynchronizer([algo.cbf, algo.actor])
+    if dtype == "bf16":
+        from gcbf_amd.utils.amp import enable_bf16
+        enable_bf16(algo)
+
+    total_schedule = 500_000  # exploration schedule of the paper config
+    data = env.reset()
+
+    def one_step(step):
+        nonlocal data
+        data.update(u_ref=env.u_ref(data))
+        action = algo.step(data, prob=1 - (step - 1) / total_schedule)
+        next_data, reward, done, info = env.step(action)
+        data = env.reset() if done else next_data
+        if algo.is_update(step):
+            algo.update(step, None)
+
+    # ---- warmup (untimed) ----
+    for step in range(1, args.warmup + 1):
+        one_step(step)
+
+    # ---- timed region ----
+    _barrier_sync(device)
+    t0 = time.perf_counter()
+    for step in range(args.warmup + 1, args.warmup + args.steps + 1):
+        one_step(step)
+    _barrier_sync(device)
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if torch.distributed.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        value = args.gpus * args.steps / elapsed
+        print(json.dumps({
+            "metric": "env-steps/sec (whole node)",
+            "value": round(value, 2),
+            "unit": "env-steps/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "GCBF (CBFGNN 12.2M + GNNController 12.2M)",
+                "env": args.env,
+                "num_agents": args.num_agents,
+                "num_obs": args.obs,
+                "global_batch": args.batch_size * args.gpus,
+                "seq_len": None,
+                "parallelism": f"dp{args.gpus}",
+            },
+        }), flush=True)
+    cleanup_distributed()
+
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=1024)
+    p.add_argument("--warmup", type=int, default=576)
+    p.add_argument("--env", type=str, default="DubinsCar")
+    p.add_argument("-n", "--num-agents", type=int, default=16)
+    p.add_argument("--obs", type=int, default=0)
+    p.add_argument("--batch-size", type=int, default=512)
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--dtype", type=str, default=None,
+                   choices=[None, "bf16", "fp32"])
+    run(p.parse_args())

@@ -1,0 +1,94 @@
+"""Replay buffer of graphs with balanced safe/unsafe segment sampling.
+
+Behavioral equivalent of the reference ``Buffer`` (gcbf/algo/buffer.py:11-95):
+graphs are stored as a list (on device — 288 GB HBM3E holds the full 100k-graph
+window comfortably), with index lists of safe/unsafe steps; sampling draws
+balanced safe/unsafe indices and expands each into a symmetric window of
+consecutive steps, deduplicated by clamping to the previous window's end.
+"""
+from __future__ import annotations
+
+import random
+from typing import List
+
+import numpy as np
+
+from ..graph import GraphBatch
+
+
+class Buffer:
+    MAX_SIZE = 100000
+
+    def __init__(self):
+        self._data: List[GraphBatch] = []
+        self.safe_data: List[int] = []
+        self.unsafe_data: List[int] = []
+
+    @property
+    def data(self) -> List[GraphBatch]:
+        return self._data
+
+    @property
+    def size(self) -> int:
+        return len(self._data)
+
+    def append(self, data: GraphBatch, is_safe: bool):
+        self._data.append(data)
+        (self.safe_data if is_safe else self.unsafe_data).append(self.size - 1)
+        if self.size > self.MAX_SIZE:
+            del self._data[0]
+            if 0 in self.safe_data:
+                self.safe_data.remove(0)
+            else:
+                self.unsafe_data.remove(0)
+            self.safe_data = [i - 1 for i in self.safe_data]
+            self.unsafe_data = [i - 1 for i in self.unsafe_data]
+
+    def merge(self, other: "Buffer"):
+        size_init = self.size
+        self._data += other.data
+        self.safe_data.extend(i + size_init for i in other.safe_data)
+        self.unsafe_data.extend(i + size_init for i in other.unsafe_data)
+        if self.size > self.MAX_SIZE:
+            overflow = self.size - self.MAX_SIZE
+            for i in range(overflow):
+                if i in self.safe_data:
+                    self.safe_data.remove(i)
+                else:
+                    self.unsafe_data.remove(i)
+            self.safe_data = [i - overflow for i in self.safe_data]
+            self.unsafe_data = [i - overflow for i in self.unsafe_data]
+            del self._data[:overflow]
+
+    def clear(self):
+        self._data.clear()
+        self.safe_data = []
+        self.unsafe_data = []
+
+    def sample(self, n: int, m: int = 1,
+               balanced_sampling: bool = False) -> List[GraphBatch]:
+        """Sample n trajectory segments of up to m consecutive graphs.
+
+        Matches reference gcbf/algo/buffer.py:61-95: indices are drawn (50/50
+        from safe/unsafe lists when balanced), sorted, and each expanded to
+        [i - m//2, i + m//2] with the lower bound clamped to the previous
+        segment's upper bound (avoids duplicated graphs).
+        """
+        assert self.size >= max(n, m)
+        if not balanced_sampling:
+            index = np.sort(np.random.randint(0, self.size, n))
+        else:
+            index_unsafe, index_safe = [], []
+            if self.unsafe_data:
+                index_unsafe = random.choices(self.unsafe_data, k=n // 2)
+            if self.safe_data:
+                index_safe = random.choices(self.safe_data, k=n // 2)
+            index = sorted(index_safe + index_unsafe)
+
+        data_list: List[GraphBatch] = []
+        ub = 0
+        for i in index:
+            lb = max(i - m // 2, ub)
+            ub = min(i + m // 2 + 1, self.size)
+            data_list.extend(self._data[lb:ub])
+        return data_list

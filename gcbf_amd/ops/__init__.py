@@ -1,0 +1,165 @@
+"""Op dispatch layer: HIP/CDNA4 kernels on GPU, eager PyTorch on CPU.
+
+Every hot-path op has (a) an eager PyTorch implementation in ``eager.py``
+(the numerics oracle, used on CPU and in tests) and (b) a HIP kernel for
+gfx950 in ``gcbf_amd/ops/hip`` exposed through the in-tree extension
+``gcbf_amd._C``.  On a GPU tensor the HIP path is mandatory: a missing
+extension raises instead of silently falling back to eager (so GPU runs
+always exercise the native kernels).  Set ``GCBF_AMD_ALLOW_EAGER_GPU=1``
+to permit the eager path on GPU (bring-up/debug only).
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+from torch import Tensor
+
+from . import eager
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from gcbf_amd import _C  # built in-tree by setup.py / __graft_entry__.build()
+        _EXT = _C
+    except ImportError as e:  # remember why, report on first GPU use
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def hip_available() -> bool:
+    return _load_ext() is not None
+
+
+def _require_ext(opname: str):
+    ext = _load_ext()
+    if ext is None:
+        if os.environ.get("GCBF_AMD_ALLOW_EAGER_GPU") == "1":
+            return None
+        raise RuntimeError(
+            f"gcbf_amd HIP extension is required for {opname} on GPU but could "
+            f"not be imported ({_EXT_ERR}). Build it with "
+            f"`python setup.py build_ext --inplace` "
+            f"(or set GCBF_AMD_ALLOW_EAGER_GPU=1 to allow the slow eager path).")
+    return ext
+
+
+# --------------------------------------------------------------------------
+# segment attention aggregation (softmax over incoming edges + weighted sum)
+# --------------------------------------------------------------------------
+
+class _SegmentAttnAggregate(torch.autograd.Function):
+    """Fused scatter-softmax + weighted scatter-sum with analytic VJP.
+
+    forward: a_e = softmax over {e: dst[e]=n} of gate_e;  out_n = sum a_e m_e
+    backward: dm_e = a_e * g_{n(e)};  s_e = <m_e, g_{n(e)}>;
+              dgate_e = a_e * (s_e - sum_{e' in n} a_e' s_e')
+    """
+
+    @staticmethod
+    def forward(ctx, msg: Tensor, gate: Tensor, dst: Tensor, num_nodes: int):
+        ext = _require_ext("segment_attn_aggregate") if msg.is_cuda else None
+        if ext is not None and msg.is_cuda:
+            att, out = ext.segment_attn_fwd(msg, gate, dst, num_nodes)
+        else:
+            att = eager.segment_softmax(gate, dst, num_nodes)
+            out = eager.segment_sum(att * msg, dst, num_nodes)
+        ctx.save_for_backward(msg, att, dst)
+        ctx.num_nodes = num_nodes
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out: Tensor):
+        msg, att, dst = ctx.saved_tensors
+        n = ctx.num_nodes
+        ext = _EXT
+        if ext is not None and msg.is_cuda:
+            dmsg, dgate = ext.segment_attn_bwd(grad_out.contiguous(), msg, att,
+                                               dst, n)
+        else:
+            g = grad_out.index_select(0, dst)            # (E, D)
+            dmsg = att * g
+            s = (msg * g).sum(dim=1, keepdim=True)       # (E, 1)
+            seg = torch.zeros(n, 1, dtype=s.dtype, device=s.device)
+            seg = seg.index_add(0, dst, att * s)
+            dgate = att * (s - seg.index_select(0, dst))
+        return dmsg, dgate, None, None
+
+
+def segment_attn_aggregate(msg: Tensor, gate: Tensor, dst: Tensor,
+                           num_nodes: int) -> Tensor:
+    if not msg.is_cuda:
+        return eager.segment_attn_aggregate(msg, gate, dst, num_nodes)
+    return _SegmentAttnAggregate.apply(msg, gate, dst, num_nodes)
+
+
+def segment_softmax(gate: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
+    return eager.segment_softmax(gate, dst, num_nodes)
+
+
+class _SegmentMax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, values: Tensor, dst: Tensor, num_nodes: int):
+        out = eager.segment_max(values.detach(), dst, num_nodes)
+        # argmax for backward: first edge attaining the max in its segment
+        sel = out.index_select(0, dst)
+        is_max = (values.detach() == sel)
+        # keep only the first maximal edge per (segment, feature)
+        order = torch.arange(values.shape[0], device=values.device)
+        big = values.shape[0] + 1
+        cand = torch.where(is_max, order.unsqueeze(1), torch.full_like(
+            is_max, big, dtype=torch.long))
+        first = torch.full((num_nodes, values.shape[1]), big, dtype=torch.long,
+                           device=values.device)
+        first = first.scatter_reduce(0, dst.unsqueeze(1).expand_as(cand), cand,
+                                     reduce="amin", include_self=True)
+        ctx.save_for_backward(dst, first)
+        ctx.E = values.shape[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out: Tensor):
+        dst, first = ctx.saved_tensors
+        E = ctx.E
+        dval = torch.zeros(E + 1, grad_out.shape[1], dtype=grad_out.dtype,
+                           device=grad_out.device)
+        idx = torch.clamp(first, max=E)
+        dval.scatter_add_(0, idx, grad_out)
+        return dval[:E], None, None
+
+
+def segment_max(values: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
+    return _SegmentMax.apply(values, dst, num_nodes)
+
+
+# --------------------------------------------------------------------------
+# graph construction
+# --------------------------------------------------------------------------
+
+def dense_radius_graph(pos: Tensor, agent_mask: Optional[Tensor],
+                       comm_radius: float,
+                       max_neighbors: Optional[int] = None,
+                       batch: int = 1) -> Tensor:
+    if pos.is_cuda:
+        ext = _require_ext("dense_radius_graph")
+        if ext is not None:
+            n_rec = (pos.shape[0] // batch if agent_mask is None
+                     else int(agent_mask.view(batch, -1)[0].sum().item()))
+            return ext.dense_radius_graph(
+                pos.contiguous(), batch, n_rec, float(comm_radius),
+                -1 if max_neighbors is None else int(max_neighbors))
+    return eager.dense_radius_graph(pos, agent_mask, comm_radius,
+                                    max_neighbors, batch)
+
+
+def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],
+                         batch: int, pos_dim: int, diag_offset: float) -> Tensor:
+    return eager.pairwise_dist_masked(states, agent_mask, batch, pos_dim,
+                                      diag_offset)
