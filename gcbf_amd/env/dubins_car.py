@@ -243,8 +243,9 @@ class DubinsCar(SimpleCar):
             edge_info.index_select(0, edge_index[1])
 
     # dense builder with agent-only receivers comes from SimpleCar's
-    # add_communication_links via ops.dense_radius_graph (the agent_mask on
-    # DubinsCar graphs restricts receivers to agents)
+    # add_communication_links; the fused GPU kernel computes the 5-dim
+    # dubins edge_attr in its fill pass
+    _attr_kind = ops.ATTR_DUBINS
 
     @property
     def state_lim(self) -> Tuple[Tensor, Tensor]:

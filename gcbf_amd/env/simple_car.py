@@ -178,12 +178,16 @@ class SimpleCar(MultiAgentEnv):
         return state.index_select(0, edge_index[0]) - \
             state.index_select(0, edge_index[1])
 
+    # edge_attr kind for the fused GPU builder (ops/hip/graph_build.hip)
+    _attr_kind = ops.ATTR_DIFF
+
     def add_communication_links(self, data: GraphBatch) -> GraphBatch:
-        edge_index = ops.dense_radius_graph(
-            data.pos, data.agent_mask, self._params["comm_radius"],
-            self._max_neighbors, batch=data.num_graphs)
-        data.update(edge_index=edge_index,
-                    edge_attr=self.edge_attr(data.states, edge_index))
+        n_rec = None if data.agent_mask is None else self.num_agents
+        edge_index, edge_attr = ops.build_graph(
+            data.pos, data.states, n_rec, self._params["comm_radius"],
+            self._max_neighbors, data.num_graphs, self._attr_kind,
+            self.edge_dim, self.edge_attr)
+        data.update(edge_index=edge_index, edge_attr=edge_attr)
         return data
 
     # uniform node counts make the batched rebuild identical to the per-graph

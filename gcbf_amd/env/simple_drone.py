@@ -213,6 +213,8 @@ class SimpleDrone(MultiAgentEnv):
         return state.index_select(0, edge_index[0]) - \
             state.index_select(0, edge_index[1])
 
+    from .. import ops as _ops
+    _attr_kind = _ops.ATTR_DIFF
     add_communication_links = SimpleCar.add_communication_links
     add_communication_links_batched = SimpleCar.add_communication_links
 

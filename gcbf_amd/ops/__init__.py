@@ -51,6 +51,13 @@ def _require_ext(opname: str):
     return ext
 
 
+def _csr_ptr(dst: Tensor, num_nodes: int) -> Tensor:
+    """CSR segment pointer from a dst-sorted edge list (device op, no host
+    sync)."""
+    bounds = torch.arange(num_nodes + 1, device=dst.device, dtype=dst.dtype)
+    return torch.searchsorted(dst, bounds).to(torch.int32)
+
+
 # --------------------------------------------------------------------------
 # segment attention aggregation (softmax over incoming edges + weighted sum)
 # --------------------------------------------------------------------------
@@ -67,10 +74,15 @@ class _SegmentAttnAggregate(torch.autograd.Function):
     def forward(ctx, msg: Tensor, gate: Tensor, dst: Tensor, num_nodes: int):
         ext = _require_ext("segment_attn_aggregate") if msg.is_cuda else None
         if ext is not None and msg.is_cuda:
-            att, out = ext.segment_attn_fwd(msg, gate, dst, num_nodes)
+            ptr = _csr_ptr(dst, num_nodes)
+            att, out = ext.segment_attn_fwd(
+                msg.contiguous(), gate.reshape(-1).contiguous(), ptr)
+            att = att.unsqueeze(-1)
+            ctx.ptr = ptr
         else:
             att = eager.segment_softmax(gate, dst, num_nodes)
             out = eager.segment_sum(att * msg, dst, num_nodes)
+            ctx.ptr = None
         ctx.save_for_backward(msg, att, dst)
         ctx.num_nodes = num_nodes
         return out
@@ -80,9 +92,11 @@ class _SegmentAttnAggregate(torch.autograd.Function):
         msg, att, dst = ctx.saved_tensors
         n = ctx.num_nodes
         ext = _EXT
-        if ext is not None and msg.is_cuda:
-            dmsg, dgate = ext.segment_attn_bwd(grad_out.contiguous(), msg, att,
-                                               dst, n)
+        if ext is not None and msg.is_cuda and ctx.ptr is not None:
+            dmsg, dgate = ext.segment_attn_bwd(
+                grad_out.contiguous(), msg.contiguous(),
+                att.reshape(-1).contiguous(), ctx.ptr)
+            dgate = dgate.unsqueeze(-1)
         else:
             g = grad_out.index_select(0, dst)            # (E, D)
             dmsg = att * g
@@ -147,16 +161,46 @@ def dense_radius_graph(pos: Tensor, agent_mask: Optional[Tensor],
                        comm_radius: float,
                        max_neighbors: Optional[int] = None,
                        batch: int = 1) -> Tensor:
-    if pos.is_cuda:
-        ext = _require_ext("dense_radius_graph")
-        if ext is not None:
-            n_rec = (pos.shape[0] // batch if agent_mask is None
-                     else int(agent_mask.view(batch, -1)[0].sum().item()))
-            return ext.dense_radius_graph(
-                pos.contiguous(), batch, n_rec, float(comm_radius),
-                -1 if max_neighbors is None else int(max_neighbors))
+    """Edge list only (eager helper; GPU paths use :func:`build_graph`)."""
     return eager.dense_radius_graph(pos, agent_mask, comm_radius,
                                     max_neighbors, batch)
+
+
+# edge_attr kinds understood by the fused builder (mirrors graph_build.hip)
+ATTR_DIFF = 0    # states[src] - states[dst]
+ATTR_DUBINS = 1  # [x, y, theta, v cos, v sin] difference
+
+
+def build_graph(pos: Tensor, states: Tensor, n_rec: Optional[int],
+                comm_radius: float, max_neighbors: Optional[int],
+                batch: int, attr_kind: int, attr_dim: int,
+                eager_attr_fn) -> tuple:
+    """Fused radius-graph + edge_attr construction.
+
+    ``n_rec`` is the number of receiver (agent) nodes per graph — agents are
+    the first ``n_rec`` rows of each graph; ``None`` means every node
+    receives.  GPU: one count/scan/fill kernel pair writing edge_index and
+    edge_attr in a single pass.  CPU: eager builder + the env's edge_attr
+    function.  Returns (edge_index (2,E) long, edge_attr (E, attr_dim)).
+    """
+    N = pos.shape[0] // batch
+    if n_rec is None:
+        n_rec = N
+    if pos.is_cuda:
+        ext = _require_ext("build_graph")
+        if ext is not None:
+            topk = -1 if max_neighbors is None else int(max_neighbors)
+            ei, ea = ext.build_graph(
+                pos.contiguous(), states.contiguous(), batch, n_rec,
+                float(comm_radius), topk, int(attr_kind), int(attr_dim))
+            return ei, ea
+    am = None
+    if n_rec != N:
+        am = torch.zeros(batch, N, dtype=torch.bool, device=pos.device)
+        am[:, :n_rec] = True
+        am = am.view(-1)
+    ei = eager.dense_radius_graph(pos, am, comm_radius, max_neighbors, batch)
+    return ei, eager_attr_fn(states, ei)
 
 
 def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],

@@ -93,8 +93,10 @@ def dense_radius_graph(
     B = batch
     N = pos.shape[0] // B
     p = pos.view(B, N, -1)
-    # (B, N, N) pairwise distances: dist[b, i, j] = |p_i - p_j|
-    dist = torch.cdist(p, p)
+    # (B, N, N) pairwise distances dist[b, i, j] = |p_i - p_j| via explicit
+    # differences (same rounding as the reference's torch.norm and the HIP
+    # kernel; cdist's matmul trick rounds differently at the radius boundary)
+    dist = (p.unsqueeze(2) - p.unsqueeze(1)).norm(dim=-1)
     big = comm_radius + 1.0
     eye = torch.eye(N, device=pos.device, dtype=dist.dtype)
     if agent_mask is None:
@@ -132,7 +134,7 @@ def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],
     B = batch
     N = states.shape[0] // B
     p = states.view(B, N, -1)[..., :pos_dim]
-    dist = torch.cdist(p, p)
+    dist = (p.unsqueeze(2) - p.unsqueeze(1)).norm(dim=-1)
     eye = torch.eye(N, device=states.device, dtype=dist.dtype)
     if agent_mask is None:
         return dist + eye * diag_offset

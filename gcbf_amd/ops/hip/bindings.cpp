@@ -1,0 +1,115 @@
+// Python bindings for the gcbf_amd HIP/CDNA4 kernels (gfx950).
+//
+// Thin torch glue: tensor checks + allocation here, all device code in the
+// .hip translation units (compiled with --offload-arch=gfx950).
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+extern "C" void launch_seg_attn_fwd(const float* msg, const float* gate,
+                                    const int* ptr, float* att, float* out,
+                                    int N, int D, hipStream_t stream);
+extern "C" void launch_seg_attn_bwd(const float* grad_out, const float* msg,
+                                    const float* att, const int* ptr,
+                                    float* dmsg, float* dgate, int N, int D,
+                                    hipStream_t stream);
+extern "C" void launch_radius_count(const float* pos, int* counts, int B,
+                                    int N, int n_rec, int P, float r,
+                                    int topk, hipStream_t stream);
+extern "C" void launch_radius_fill(const float* pos, const float* states,
+                                   const int* offsets, long* edge_index,
+                                   float* edge_attr, long E_total, int B,
+                                   int N, int n_rec, int P, int S, int A,
+                                   float r, int topk, int attr_kind,
+                                   hipStream_t stream);
+
+namespace {
+
+#define CHECK_IN(x)                                                    \
+    TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                    \
+    TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+hipStream_t current_stream() {
+    return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+std::vector<torch::Tensor> segment_attn_fwd(torch::Tensor msg,
+                                            torch::Tensor gate,
+                                            torch::Tensor ptr) {
+    CHECK_IN(msg);
+    CHECK_IN(gate);
+    CHECK_IN(ptr);
+    TORCH_CHECK(msg.scalar_type() == torch::kFloat32, "msg must be fp32");
+    TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+    const int64_t E = msg.size(0), D = msg.size(1);
+    const int64_t N = ptr.size(0) - 1;
+    auto att = torch::empty({E}, msg.options());
+    auto out = torch::empty({N, D}, msg.options());
+    if (N > 0)
+        launch_seg_attn_fwd(msg.data_ptr<float>(), gate.data_ptr<float>(),
+                            ptr.data_ptr<int>(), att.data_ptr<float>(),
+                            out.data_ptr<float>(), (int)N, (int)D,
+                            current_stream());
+    return {att, out};
+}
+
+std::vector<torch::Tensor> segment_attn_bwd(torch::Tensor grad_out,
+                                            torch::Tensor msg,
+                                            torch::Tensor att,
+                                            torch::Tensor ptr) {
+    CHECK_IN(grad_out);
+    CHECK_IN(msg);
+    CHECK_IN(att);
+    CHECK_IN(ptr);
+    const int64_t E = msg.size(0), D = msg.size(1);
+    const int64_t N = ptr.size(0) - 1;
+    auto dmsg = torch::empty_like(msg);
+    auto dgate = torch::empty({E}, msg.options());
+    if (N > 0)
+        launch_seg_attn_bwd(grad_out.data_ptr<float>(), msg.data_ptr<float>(),
+                            att.data_ptr<float>(), ptr.data_ptr<int>(),
+                            dmsg.data_ptr<float>(), dgate.data_ptr<float>(),
+                            (int)N, (int)D, current_stream());
+    return {dmsg, dgate};
+}
+
+std::vector<torch::Tensor> build_graph(torch::Tensor pos,
+                                       torch::Tensor states, int64_t B,
+                                       int64_t n_rec, double r, int64_t topk,
+                                       int64_t attr_kind, int64_t attr_dim) {
+    CHECK_IN(pos);
+    CHECK_IN(states);
+    const int64_t N = pos.size(0) / B;
+    const int64_t P = pos.size(1), S = states.size(1);
+    auto stream = current_stream();
+    auto counts = torch::empty({B * n_rec},
+                               pos.options().dtype(torch::kInt32));
+    launch_radius_count(pos.data_ptr<float>(), counts.data_ptr<int>(),
+                        (int)B, (int)N, (int)n_rec, (int)P, (float)r,
+                        (int)topk, stream);
+    auto incl = counts.cumsum(0, torch::kInt32);
+    auto offsets = (incl - counts).contiguous();
+    const int64_t E = incl.numel() ? incl[-1].item<int64_t>() : 0;
+    auto edge_index = torch::empty({2, E},
+                                   pos.options().dtype(torch::kInt64));
+    auto edge_attr = torch::empty({E, attr_dim}, pos.options());
+    if (E > 0)
+        launch_radius_fill(pos.data_ptr<float>(), states.data_ptr<float>(),
+                           offsets.data_ptr<int>(),
+                           edge_index.data_ptr<int64_t>(),
+                           edge_attr.data_ptr<float>(), E, (int)B, (int)N,
+                           (int)n_rec, (int)P, (int)S, (int)attr_dim,
+                           (float)r, (int)topk, (int)attr_kind, stream);
+    return {edge_index, edge_attr};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("segment_attn_fwd", &segment_attn_fwd,
+          "fused scatter-softmax + weighted scatter-sum (forward)");
+    m.def("segment_attn_bwd", &segment_attn_bwd,
+          "fused scatter-softmax + weighted scatter-sum (backward)");
+    m.def("build_graph", &build_graph,
+          "batched dense radius graph + edge_attr (count/scan/fill)");
+}
