@@ -70,15 +70,28 @@ def run(args):
 
     total_schedule = 500_000  # exploration schedule of the paper config
     data = env.reset()
+    prof = {"rollout_s": 0.0, "update_s": 0.0, "updates": 0}
 
-    def one_step(step):
+    def one_step(step, timed=False):
         nonlocal data
+        t0 = time.perf_counter() if timed else 0.0
         data.update(u_ref=env.u_ref(data))
         action = algo.step(data, prob=1 - (step - 1) / total_schedule)
         next_data, reward, done, info = env.step(action)
         data = env.reset() if done else next_data
         if algo.is_update(step):
+            if timed and use_cuda:
+                torch.cuda.synchronize(device)
+                t1 = time.perf_counter()
+                prof["rollout_s"] += t1 - t0
+                algo.update(step, None)
+                torch.cuda.synchronize(device)
+                prof["update_s"] += time.perf_counter() - t1
+                prof["updates"] += 1
+                return
             algo.update(step, None)
+        if timed:
+            prof["rollout_s"] += time.perf_counter() - t0
 
     # ---- warmup (untimed) ----
     for step in range(1, args.warmup + 1):
@@ -88,9 +101,16 @@ def run(args):
     _barrier_sync(device)
     t0 = time.perf_counter()
     for step in range(args.warmup + 1, args.warmup + args.steps + 1):
-        one_step(step)
+        one_step(step, timed=args.profile)
     _barrier_sync(device)
     elapsed = time.perf_counter() - t0
+    if args.profile and rank == 0:
+        import sys
+        print(f"# profile: rollout {prof['rollout_s']:.2f}s "
+              f"({prof['rollout_s'] / max(args.steps - prof['updates'], 1) * 1e3:.2f} ms/step), "
+              f"update {prof['update_s']:.2f}s over {prof['updates']} updates "
+              f"({prof['update_s'] / max(prof['updates'], 1):.2f} s/update)",
+              file=sys.stderr, flush=True)
 
     # max over ranks
     if torch.distributed.is_initialized():
@@ -139,4 +159,6 @@ if __name__ == "__main__":
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--dtype", type=str, default=None,
                    choices=[None, "bf16", "fp32"])
+    p.add_argument("--profile", action="store_true", default=False,
+                   help="print rollout/update time split to stderr")
     run(p.parse_args())

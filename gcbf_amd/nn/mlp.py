@@ -3,8 +3,16 @@
 Mirrors the reference MLP (gcbf/nn/mlp.py:9-47) including its state-dict
 layout: parameters live in an ``nn.Sequential`` named ``net`` with Linear
 modules at the same indices (activations between), and ``limit_lip=True``
-wraps each Linear in old-style spectral norm (keys ``weight_orig``,
+uses spectral normalization with old-style keys (``weight_orig``,
 ``weight_u``, ``weight_v``) — so reference checkpoints load directly.
+
+Spectral norm is implemented natively (:class:`SNLinear`) rather than via
+``torch.nn.utils.spectral_norm`` because the stock hook's power iteration
+uses ``normalize(..., out=buffer)`` which breaks under bf16 autocast; here
+the power iteration runs in fp32 with autocast disabled, while the actual
+GEMM autocasts to bf16 on GPU.  Semantics match torch's: one power
+iteration per forward in training mode, σ = uᵀWv with u,v as constants,
+effective weight W/σ.
 """
 from __future__ import annotations
 
@@ -12,7 +20,7 @@ from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn
-from torch.nn.utils import spectral_norm
+import torch.nn.functional as F
 
 
 def init_param(module: nn.Module, gain: float = 1.0) -> nn.Module:
@@ -20,6 +28,61 @@ def init_param(module: nn.Module, gain: float = 1.0) -> nn.Module:
     nn.init.orthogonal_(module.weight.data, gain=gain)
     nn.init.constant_(module.bias.data, 0)
     return module
+
+
+class SNLinear(nn.Module):
+    """Linear layer with spectral normalization (Lipschitz ≤ 1).
+
+    State-dict keys match old-style ``torch.nn.utils.spectral_norm`` on an
+    ``nn.Linear``: ``weight_orig``, ``bias``, ``weight_u``, ``weight_v``.
+    """
+
+    def __init__(self, in_features: int, out_features: int,
+                 n_power_iterations: int = 1, eps: float = 1e-12):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.n_power_iterations = n_power_iterations
+        self.eps = eps
+        self.weight_orig = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features))
+        nn.init.kaiming_uniform_(self.weight_orig, a=5 ** 0.5)
+        u = F.normalize(torch.randn(out_features), dim=0, eps=eps)
+        v = F.normalize(torch.randn(in_features), dim=0, eps=eps)
+        self.register_buffer("weight_u", u)
+        self.register_buffer("weight_v", v)
+
+    @property
+    def weight(self) -> torch.Tensor:
+        """Effective (normalized) weight — read-only convenience."""
+        with torch.no_grad():
+            sigma = torch.dot(self.weight_u,
+                              torch.mv(self.weight_orig, self.weight_v))
+        return self.weight_orig / sigma
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        W = self.weight_orig
+        if self.training:
+            with torch.no_grad():
+                # fp32 power iteration regardless of autocast
+                Wf = W.detach().float()
+                u, v = self.weight_u, self.weight_v
+                for _ in range(self.n_power_iterations):
+                    v = F.normalize(Wf.t().mv(u), dim=0, eps=self.eps)
+                    u = F.normalize(Wf.mv(v), dim=0, eps=self.eps)
+                self.weight_u.copy_(u)
+                self.weight_v.copy_(v)
+        # σ differentiable through W (u, v constants) — same as torch's hook.
+        # clone() so later in-place buffer updates (next forward's power
+        # iteration) don't invalidate this graph's saved tensors.
+        u = self.weight_u.clone().to(W.dtype)
+        v = self.weight_v.clone().to(W.dtype)
+        sigma = torch.dot(u, torch.mv(W, v))
+        return F.linear(x, W / sigma, self.bias)
+
+    def extra_repr(self) -> str:
+        return f"in_features={self.in_features}, " \
+               f"out_features={self.out_features}, spectral_norm=True"
 
 
 class MLP(nn.Module):
@@ -35,11 +98,15 @@ class MLP(nn.Module):
             hidden_activation = nn.ReLU()
 
         def make_linear(n_in: int, n_out: int) -> nn.Module:
-            lin = nn.Linear(n_in, n_out)
-            if init:
-                init_param(lin, gain=gain)
             if limit_lip:
-                lin = spectral_norm(lin)
+                lin = SNLinear(n_in, n_out)
+                if init:
+                    nn.init.orthogonal_(lin.weight_orig.data, gain=gain)
+                    nn.init.constant_(lin.bias.data, 0)
+            else:
+                lin = nn.Linear(n_in, n_out)
+                if init:
+                    init_param(lin, gain=gain)
             return lin
 
         layers = []

@@ -73,10 +73,15 @@ class _SegmentAttnAggregate(torch.autograd.Function):
     @staticmethod
     def forward(ctx, msg: Tensor, gate: Tensor, dst: Tensor, num_nodes: int):
         ext = _require_ext("segment_attn_aggregate") if msg.is_cuda else None
+        ctx.msg_dtype = msg.dtype
+        ctx.gate_dtype = gate.dtype
         if ext is not None and msg.is_cuda:
+            # kernel computes fp32 (softmax stability; the op is tiny next to
+            # the bf16 GEMMs around it) — upcast bf16 inputs at the boundary
+            msg = msg.float()
             ptr = _csr_ptr(dst, num_nodes)
             att, out = ext.segment_attn_fwd(
-                msg.contiguous(), gate.reshape(-1).contiguous(), ptr)
+                msg.contiguous(), gate.reshape(-1).float().contiguous(), ptr)
             att = att.unsqueeze(-1)
             ctx.ptr = ptr
         else:
@@ -94,9 +99,11 @@ class _SegmentAttnAggregate(torch.autograd.Function):
         ext = _EXT
         if ext is not None and msg.is_cuda and ctx.ptr is not None:
             dmsg, dgate = ext.segment_attn_bwd(
-                grad_out.contiguous(), msg.contiguous(),
+                grad_out.float().contiguous(), msg.contiguous(),
                 att.reshape(-1).contiguous(), ctx.ptr)
             dgate = dgate.unsqueeze(-1)
+            dmsg = dmsg.to(ctx.msg_dtype)
+            dgate = dgate.to(ctx.gate_dtype)
         else:
             g = grad_out.index_select(0, dst)            # (E, D)
             dmsg = att * g

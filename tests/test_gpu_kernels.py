@@ -170,3 +170,30 @@ def test_gcbf_update_gpu():
         if algo.is_update(step):
             out = algo.update(step, None)
     assert all(0 <= v <= 1 for v in out.values())
+
+
+def test_gcbf_update_gpu_bf16():
+    """bf16 autocast path: SNLinear power iteration + segment kernels must
+    cope with mixed dtypes."""
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.trainer.utils import set_seed
+    from gcbf_amd.utils.amp import enable_bf16
+    set_seed(0)
+    dev = torch.device("cuda")
+    env = make_env("DubinsCar", 16, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=32)
+    enable_bf16(algo)
+    data = env.reset()
+    for step in range(1, 33):
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.5)
+        assert a.dtype == torch.float32  # module boundary casts back
+        data, r, done, info = env.step(a)
+        if done:
+            data = env.reset()
+        if algo.is_update(step):
+            out = algo.update(step, None)
+    assert all(0 <= v <= 1 for v in out.values())
