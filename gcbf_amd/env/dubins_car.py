@@ -311,10 +311,15 @@ class DubinsCar(SimpleCar):
         eye = torch.eye(N, device=data.device, dtype=dist.dtype)[:n]
         return pd, dist + eye * diag_offset
 
+    _env_kind = ops.ENV_DUBINS
+
     def safe_mask(self, data: GraphBatch, return_edge: bool = False) -> Tensor:
         r = self._params["car_radius"]
         if return_edge:
             return data.edge_attr[:, :2].norm(dim=-1) > 4 * r
+        m = self._fused_mask(data, "safe")
+        if m is not None:
+            return m
         # diag offset 4r+1, threshold 3r (reference dubins_car.py:835-838)
         _, dist = self._pairwise_agent_rows(data, 4 * r + 1)
         return (dist > 3 * r).min(dim=2)[0].reshape(-1).bool()
@@ -323,6 +328,9 @@ class DubinsCar(SimpleCar):
         r = self._params["car_radius"]
         if return_edge:
             return data.edge_attr[:, :2].norm(dim=-1) < 2 * r
+        m = self._fused_mask(data, "unsafe")
+        if m is not None:
+            return m
         pd, dist = self._pairwise_agent_rows(data, 4 * r + 1)
         collision = (dist < 2 * r).max(dim=2)[0]
 
@@ -343,6 +351,9 @@ class DubinsCar(SimpleCar):
         r = self._params["car_radius"]
         if self._mode not in ("train", "test", "demo_1", "demo_2"):
             raise NotImplementedError
+        m = self._fused_mask(data, "collision")
+        if m is not None:
+            return m
         _, dist = self._pairwise_agent_rows(data, 2 * r + 1)
         return (dist < 2 * r).max(dim=2)[0].reshape(-1).bool()
 

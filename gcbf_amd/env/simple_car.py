@@ -256,10 +256,28 @@ class SimpleCar(MultiAgentEnv):
         eye = torch.eye(N, device=data.device, dtype=dist.dtype)
         return pd, dist + eye * diag_offset
 
+    _env_kind = ops.ENV_CAR
+
+    def _fused_mask(self, data: GraphBatch, which: str):
+        return ops.fused_masks(data.states, data.num_graphs,
+                               self._mask_rows(data), self._params.get(
+                                   "car_radius",
+                                   self._params.get("drone_radius")),
+                               self._env_kind, which)
+
+    def _mask_rows(self, data: GraphBatch) -> int:
+        # receiver rows per graph: all nodes for SimpleCar, agents otherwise
+        if data.agent_mask is None:
+            return data.nodes_per_graph
+        return self.num_agents
+
     def safe_mask(self, data: GraphBatch, return_edge: bool = False) -> Tensor:
         r = self._params["car_radius"]
         if return_edge:
             return data.edge_attr[:, :2].norm(dim=-1) > 4 * r
+        m = self._fused_mask(data, "safe")
+        if m is not None:
+            return m
         _, dist = self._pairwise(data, 4 * r + 1)
         return (dist > 4 * r).min(dim=2)[0].reshape(-1).bool()
 
@@ -267,6 +285,9 @@ class SimpleCar(MultiAgentEnv):
         r = self._params["car_radius"]
         if return_edge:
             return data.edge_attr[:, :2].norm(dim=-1) < 2 * r
+        m = self._fused_mask(data, "unsafe")
+        if m is not None:
+            return m
         pd, dist = self._pairwise(data, 4 * r + 1)
         collision = (dist < 2 * r).max(dim=2)[0]
 
@@ -285,6 +306,9 @@ class SimpleCar(MultiAgentEnv):
 
     def collision_mask(self, data: GraphBatch) -> Tensor:
         r = self._params["car_radius"]
+        m = self._fused_mask(data, "collision")
+        if m is not None:
+            return m
         _, dist = self._pairwise(data, 2 * r + 1)
         return (dist < 2 * r).max(dim=2)[0].reshape(-1).bool()
 

@@ -215,8 +215,11 @@ class SimpleDrone(MultiAgentEnv):
 
     from .. import ops as _ops
     _attr_kind = _ops.ATTR_DIFF
+    _env_kind = _ops.ENV_DRONE
     add_communication_links = SimpleCar.add_communication_links
     add_communication_links_batched = SimpleCar.add_communication_links
+    _fused_mask = SimpleCar._fused_mask
+    _mask_rows = SimpleCar._mask_rows
 
     @property
     def state_lim(self) -> Tuple[Tensor, Tensor]:
@@ -277,6 +280,9 @@ class SimpleDrone(MultiAgentEnv):
         r = self._params["drone_radius"]
         if return_edge:
             return data.edge_attr[:, :3].norm(dim=-1) > 4 * r
+        m = self._fused_mask(data, "safe")
+        if m is not None:
+            return m
         _, dist = self._pairwise_agent_rows(data, 4 * r + 1)
         return (dist > 4 * r).min(dim=2)[0].reshape(-1).bool()
 
@@ -284,6 +290,9 @@ class SimpleDrone(MultiAgentEnv):
         r = self._params["drone_radius"]
         if return_edge:
             return data.edge_attr[:, :3].norm(dim=-1) < 2 * r
+        m = self._fused_mask(data, "unsafe")
+        if m is not None:
+            return m
         pd, dist = self._pairwise_agent_rows(data, 2 * r + 1)
         collision = (dist < 2 * r).max(dim=2)[0]
 
@@ -306,6 +315,9 @@ class SimpleDrone(MultiAgentEnv):
         r = self._params["drone_radius"]
         if self._mode not in ("train", "test", "demo_2"):
             raise NotImplementedError
+        m = self._fused_mask(data, "collision")
+        if m is not None:
+            return m
         _, dist = self._pairwise_agent_rows(data, 2 * r + 1)
         return (dist < 2 * r).max(dim=2)[0].reshape(-1).bool()
 

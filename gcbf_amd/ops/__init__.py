@@ -214,3 +214,26 @@ def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],
                          batch: int, pos_dim: int, diag_offset: float) -> Tensor:
     return eager.pairwise_dist_masked(states, agent_mask, batch, pos_dim,
                                       diag_offset)
+
+
+# env kinds understood by the fused mask kernel (mirrors masks.hip)
+ENV_CAR = 0
+ENV_DUBINS = 1
+ENV_DRONE = 2
+
+_WHICH = {"safe": 0, "unsafe": 1, "collision": 2}
+
+
+def fused_masks(states: Tensor, batch: int, n_rec: int, radius: float,
+                env_kind: int, which: str) -> Optional[Tensor]:
+    """One-pass batched agent mask on GPU; returns None on CPU (callers fall
+    back to the eager vectorized math)."""
+    if not states.is_cuda:
+        return None
+    ext = _require_ext("fused_masks")
+    if ext is None:
+        return None
+    idx = _WHICH[which]
+    res = ext.fused_masks(states.contiguous(), batch, n_rec, float(radius),
+                          env_kind, idx == 0, idx == 1, idx == 2)
+    return res[idx]

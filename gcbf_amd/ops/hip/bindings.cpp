@@ -22,6 +22,10 @@ extern "C" void launch_radius_fill(const float* pos, const float* states,
                                    int N, int n_rec, int P, int S, int A,
                                    float r, int topk, int attr_kind,
                                    hipStream_t stream);
+extern "C" void launch_fused_masks(const float* states, bool* safe,
+                                   bool* unsafe, bool* collision, int B,
+                                   int N, int n_rec, int S, float r, int kind,
+                                   hipStream_t stream);
 
 namespace {
 
@@ -103,9 +107,33 @@ std::vector<torch::Tensor> build_graph(torch::Tensor pos,
     return {edge_index, edge_attr};
 }
 
+std::vector<torch::Tensor> fused_masks(torch::Tensor states, int64_t B,
+                                       int64_t n_rec, double r, int64_t kind,
+                                       bool want_safe, bool want_unsafe,
+                                       bool want_coll) {
+    CHECK_IN(states);
+    const int64_t N = states.size(0) / B;
+    const int64_t S = states.size(1);
+    auto opts = states.options().dtype(torch::kBool);
+    auto none = torch::Tensor();
+    auto t_safe = want_safe ? torch::empty({B * n_rec}, opts) : none;
+    auto t_uns = want_unsafe ? torch::empty({B * n_rec}, opts) : none;
+    auto t_coll = want_coll ? torch::empty({B * n_rec}, opts) : none;
+    launch_fused_masks(
+        states.data_ptr<float>(),
+        want_safe ? t_safe.data_ptr<bool>() : nullptr,
+        want_unsafe ? t_uns.data_ptr<bool>() : nullptr,
+        want_coll ? t_coll.data_ptr<bool>() : nullptr,
+        (int)B, (int)N, (int)n_rec, (int)S, (float)r, (int)kind,
+        current_stream());
+    return {t_safe, t_uns, t_coll};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("fused_masks", &fused_masks,
+          "batched safe/unsafe/collision agent masks in one pass");
     m.def("segment_attn_fwd", &segment_attn_fwd,
           "fused scatter-softmax + weighted scatter-sum (forward)");
     m.def("segment_attn_bwd", &segment_attn_bwd,

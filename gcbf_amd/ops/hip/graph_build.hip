@@ -112,8 +112,8 @@ __device__ __forceinline__ void write_attr(float* __restrict__ attr,
         attr[0] = ss[0] - sd[0];
         attr[1] = ss[1] - sd[1];
         attr[2] = ss[2] - sd[2];
-        attr[3] = ss[3] * __cosf(ss[2]) - sd[3] * __cosf(sd[2]);
-        attr[4] = ss[3] * __sinf(ss[2]) - sd[3] * __sinf(sd[2]);
+        attr[3] = ss[3] * cosf(ss[2]) - sd[3] * cosf(sd[2]);
+        attr[4] = ss[3] * sinf(ss[2]) - sd[3] * sinf(sd[2]);
     }
 }
 

@@ -18,11 +18,9 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="gcbf_amd._C",
-            sources=[
-                os.path.join(HIP_DIR, "bindings.cpp"),
-                os.path.join(HIP_DIR, "segops.hip"),
-                os.path.join(HIP_DIR, "graph_build.hip"),
-            ],
+            sources=sorted(
+                os.path.join(HIP_DIR, f) for f in os.listdir(HIP_DIR)
+                if f.endswith((".cpp", ".hip"))),
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
