@@ -92,3 +92,56 @@ class Buffer:
             ub = min(i + m // 2 + 1, self.size)
             data_list.extend(self._data[lb:ub])
         return data_list
+
+
+class RolloutBuffer:
+    """RL-style (state, action, reward, done, log_pi, next_state) ring
+    buffer.  API-parity port of the reference's unused RolloutBuffer
+    (gcbf/algo/buffer.py:98-204); kept for downstream RL extensions."""
+
+    def __init__(self, num_agents: int, buffer_size: int, action_dim: int,
+                 device):
+        import torch
+        self._n = 0
+        self._p = 0
+        self.device = device
+        self.buffer_size = buffer_size
+        self.num_agents = num_agents
+        self.data: List[GraphBatch] = [None] * buffer_size
+        self.next_data: List[GraphBatch] = [None] * buffer_size
+        self.actions = torch.empty(buffer_size, num_agents, action_dim,
+                                   dtype=torch.float, device=device)
+        self.rewards = torch.empty(buffer_size, num_agents,
+                                   dtype=torch.float, device=device)
+        self.dones = torch.empty(buffer_size, 1, dtype=torch.float,
+                                 device=device)
+        self.log_pis = torch.empty(buffer_size, num_agents,
+                                   dtype=torch.float, device=device)
+
+    def append(self, data: GraphBatch, action, reward, done: bool, log_pi,
+               next_data: GraphBatch):
+        import torch
+        if action.ndim == 3:
+            action = action.squeeze(0)
+        self.data[self._p] = data.replace()
+        self.actions[self._p].copy_(action)
+        self.rewards[self._p].copy_(
+            torch.as_tensor(reward, device=self.device))
+        self.dones[self._p] = float(done)
+        self.log_pis[self._p].copy_(
+            torch.as_tensor(log_pi, device=self.device))
+        self.next_data[self._p] = next_data.replace()
+        self._p = (self._p + 1) % self.buffer_size
+        self._n = min(self._n + 1, self.buffer_size)
+
+    def get(self):
+        assert self._p % self.buffer_size == 0
+        idx = slice(0, self.buffer_size)
+        return (self.data[idx], self.actions[idx], self.rewards[idx],
+                self.dones[idx], self.log_pis[idx], self.next_data[idx])
+
+    def sample(self, batch_size: int):
+        idxes = np.random.randint(low=0, high=self._n, size=batch_size)
+        return ([self.data[i] for i in idxes], self.actions[idxes],
+                self.rewards[idxes], self.dones[idxes], self.log_pis[idxes],
+                [self.next_data[i] for i in idxes])

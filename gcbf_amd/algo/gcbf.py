@@ -29,6 +29,7 @@ from ..controller import GNNController
 from ..env import MultiAgentEnv
 from ..graph import GraphBatch
 from ..nn import MLP, CBFGNNLayer
+from ..utils.trace import trace_range
 from .base import Algorithm
 from .buffer import Buffer
 
@@ -139,8 +140,9 @@ class GCBF(Algorithm):
 
             graphs = GraphBatch.from_list(graph_list)
             graphs.edge_attr.requires_grad_(True)
-            h = self.cbf(graphs)
-            actions = self.actor(graphs)
+            with trace_range("gcbf/forward"):
+                h = self.cbf(graphs)
+                actions = self.actor(graphs)
 
             # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177)
             unsafe_mask = self._env.unsafe_mask(graphs)
@@ -168,13 +170,14 @@ class GCBF(Algorithm):
             # (reference gcbf/algo/gcbf.py:191-209): the VALUE reflects the
             # re-linked next graph, the GRADIENT flows through the
             # fixed-topology path.
-            graphs_next = self._env.forward_graph(graphs, actions)
-            h_next = self.cbf(graphs_next)
-            with torch.no_grad():
-                relinked = self._env.add_communication_links_batched(
-                    self._env.forward_graph(graphs, actions.detach())
-                    .detach())
-                h_next_new_link = self.cbf(relinked)
+            with trace_range("gcbf/h_dot"):
+                graphs_next = self._env.forward_graph(graphs, actions)
+                h_next = self.cbf(graphs_next)
+                with torch.no_grad():
+                    relinked = self._env.add_communication_links_batched(
+                        self._env.forward_graph(graphs, actions.detach())
+                        .detach())
+                    h_next_new_link = self.cbf(relinked)
             h_dot = (h_next - h) / self._env.dt
             h_dot_new_link = (h_next_new_link - h) / self._env.dt
             residue = (h_dot_new_link - h_dot).detach()
@@ -193,13 +196,16 @@ class GCBF(Algorithm):
 
             self.optim_cbf.zero_grad(set_to_none=True)
             self.optim_actor.zero_grad(set_to_none=True)
-            loss.backward()
+            with trace_range("gcbf/backward"):
+                loss.backward()
             if self.grad_sync is not None:
-                self.grad_sync()
-            torch.nn.utils.clip_grad_norm_(self.cbf.parameters(), 1e-3)
-            torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
-            self.optim_cbf.step()
-            self.optim_actor.step()
+                with trace_range("gcbf/grad_allreduce"):
+                    self.grad_sync()
+            with trace_range("gcbf/optim"):
+                torch.nn.utils.clip_grad_norm_(self.cbf.parameters(), 1e-3)
+                torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
+                self.optim_cbf.step()
+                self.optim_actor.step()
 
             logs.append(torch.stack([
                 loss_unsafe.detach(), loss_safe.detach(),
