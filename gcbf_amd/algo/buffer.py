@@ -23,6 +23,11 @@ class Buffer:
         self._data: List[GraphBatch] = []
         self.safe_data: List[int] = []
         self.unsafe_data: List[int] = []
+        # (index, 0-dim bool tensor) flags not yet classified — the per-step
+        # is_safe device→host sync is deferred and resolved in ONE transfer
+        # at sample/merge time (the reference syncs every rollout step,
+        # gcbf/algo/gcbf.py:133-137)
+        self._pending: List[tuple] = []
 
     @property
     def data(self) -> List[GraphBatch]:
@@ -32,10 +37,25 @@ class Buffer:
     def size(self) -> int:
         return len(self._data)
 
-    def append(self, data: GraphBatch, is_safe: bool):
+    def _resolve(self):
+        if not self._pending:
+            return
+        import torch
+        vals = torch.stack([t for _, t in self._pending]).cpu()
+        for (idx, _), v in zip(self._pending, vals):
+            (self.safe_data if bool(v) else self.unsafe_data).append(idx)
+        self._pending.clear()
+
+    def append(self, data: GraphBatch, is_safe):
         self._data.append(data)
-        (self.safe_data if is_safe else self.unsafe_data).append(self.size - 1)
+        import torch
+        if torch.is_tensor(is_safe):
+            self._pending.append((self.size - 1, is_safe))
+        else:
+            (self.safe_data if is_safe
+             else self.unsafe_data).append(self.size - 1)
         if self.size > self.MAX_SIZE:
+            self._resolve()
             del self._data[0]
             if 0 in self.safe_data:
                 self.safe_data.remove(0)
@@ -45,6 +65,8 @@ class Buffer:
             self.unsafe_data = [i - 1 for i in self.unsafe_data]
 
     def merge(self, other: "Buffer"):
+        other._resolve()
+        self._resolve()
         size_init = self.size
         self._data += other.data
         self.safe_data.extend(i + size_init for i in other.safe_data)
@@ -74,6 +96,7 @@ class Buffer:
         [i - m//2, i + m//2] with the lower bound clamped to the previous
         segment's upper bound (avoids duplicated graphs).
         """
+        self._resolve()
         assert self.size >= max(n, m)
         if not balanced_sampling:
             index = np.sort(np.random.randint(0, self.size, n))

@@ -112,7 +112,9 @@ class GCBF(Algorithm):
         action = self.actor(data)
         if np.random.rand() < prob:
             action = torch.zeros_like(action)
-        is_safe = not bool(torch.any(self._env.unsafe_mask(data)))
+        # is_safe stays a device tensor; the buffer classifies lazily in one
+        # batched transfer (no per-step host sync)
+        is_safe = torch.logical_not(torch.any(self._env.unsafe_mask(data)))
         self.buffer.append(data, is_safe)
         return action
 

@@ -62,7 +62,7 @@ class MACBF(GCBF):
         prob = max(prob, 0.5)
         if np.random.rand() < prob:
             action = torch.zeros_like(action)
-        is_safe = not bool(torch.any(self._env.unsafe_mask(data)))
+        is_safe = torch.logical_not(torch.any(self._env.unsafe_mask(data)))
         self.buffer.append(data, is_safe)
         return action
 

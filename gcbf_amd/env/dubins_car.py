@@ -184,11 +184,19 @@ class DubinsCar(SimpleCar):
                           agent_mask=agent_mask)
 
     # ------------------------------------------------------------------ step
-    def step(self, action: Tensor) -> Tuple[GraphBatch, np.ndarray, bool, dict]:
+    def step(self, action: Tensor) -> Tuple[GraphBatch, Tensor, bool, dict]:
         self._t += 1
+        # fused single-kernel path on GPU (ops/hip/env_step.hip)
+        out = ops.env_step_fused(
+            "dubins", self._data.states, self._goal, action, self.dt,
+            self._params["car_radius"], self._params["speed_limit"],
+            self._params["dist2goal"], 2.0)
+        if out is not None:
+            return self._finish_fused_step(out)
+
         # reference gcbf/env/dubins_car.py:522-615
         reward_action = -torch.norm(action, dim=1).sum() * 0.01
-        action = action + self.u_ref(self._data)
+        action = action + self._step_u_ref()
         lower_lim, upper_lim = self.action_lim
         action = torch.clamp(action, lower_lim, upper_lim)
         am = self._data.agent_mask
@@ -215,10 +223,9 @@ class DubinsCar(SimpleCar):
         reward_reach = (reach.int() - prev_reach.int()).int() * 10
         reward = reward_reach + reward_collision + reward_step + reward_action
 
-        safe = float(1.0 - collision.sum() / self.num_agents)
-        collision_agent = torch.where(collision > 0)[0]
-        return self.data, reward.detach().cpu().numpy(), done, {
-            "reach": reach, "collision": collision_agent, "safe": safe}
+        safe = 1.0 - collision.sum() / self.num_agents
+        return self.data, reward.detach(), done, {
+            "reach": reach, "collision": collision, "safe": safe}
 
     def forward_graph(self, data: GraphBatch, action: Tensor) -> GraphBatch:
         # reference gcbf/env/dubins_car.py:617-635

@@ -124,11 +124,49 @@ class SimpleCar(MultiAgentEnv):
         self._xy_max = xy_max + 0.5 * (max_interval - (xy_max - xy_min))
 
     # ------------------------------------------------------------------ step
-    def step(self, action: Tensor) -> Tuple[GraphBatch, np.ndarray, bool, dict]:
+    def _step_u_ref(self) -> Tensor:
+        """u_ref for the env's own step: reuse the value the trainer already
+        attached this step (same states, same goal — identical), else
+        compute (reference recomputes it each time,
+        gcbf/env/simple_car.py:151)."""
+        if self._data.u_ref is not None:
+            return self._data.u_ref
+        return self.u_ref(self._data)
+
+    def _get_K_tensor(self) -> Tensor:
+        if self._K is None:
+            self.u_ref(self._data)  # lazily builds the LQR gain
+        return self._K.contiguous()
+
+    def _finish_fused_step(self, out) -> Tuple[GraphBatch, Tensor, bool, dict]:
+        """Assemble the step return from the fused-kernel outputs."""
+        new_states, u_ref_next, reward, reach, collision = out
+        pos_dim = 3 if self.state_dim == 6 else 2
+        data = GraphBatch(
+            x=self._data.x, pos=new_states[:, :pos_dim], states=new_states,
+            agent_mask=self._data.agent_mask)
+        self._data = self.add_communication_links(data)
+        self._data.u_ref = u_ref_next
+        done = bool(self._t >= self.max_episode_steps or reach.all())
+        safe = 1.0 - collision.sum() / self.num_agents
+        return self._data, reward, done, {
+            "safe": safe, "reach": reach, "collision": collision}
+
+    def step(self, action: Tensor) -> Tuple[GraphBatch, Tensor, bool, dict]:
         self._t += 1
-        # reference gcbf/env/simple_car.py:146-176
+        # fused single-kernel path on GPU (ops/hip/env_step.hip)
+        out = ops.env_step_fused(
+            "car", self._data.states, self._goal, action,
+            self._get_K_tensor(), self.dt, self._params["car_radius"],
+            self._params["speed_limit"], self._params["dist2goal"], 10.0)
+        if out is not None:
+            return self._finish_fused_step(out)
+
+        # reference gcbf/env/simple_car.py:146-176.  Rewards/info stay on
+        # device (one host sync per step, for `done`); callers that need
+        # numpy use .cpu().numpy().
         reward_action = -torch.norm(action, dim=1) * 0.0001
-        action = action + self.u_ref(self._data)
+        action = action + self._step_u_ref()
         lower_lim, upper_lim = self.action_lim
         action = torch.clamp(action, lower_lim, upper_lim)
         prev_reach = torch.less(
@@ -152,12 +190,10 @@ class SimpleCar(MultiAgentEnv):
         reward_collision = -collision.int() * 2
         reward_reach = (reach.int() - prev_reach.int()) * 4
         reward = reward_reach + reward_collision + reward_step + reward_action
-        reward = reward.detach().cpu().numpy()
 
-        safe = float(1.0 - collision.sum() / self.num_agents)
-        collision_agent = torch.where(collision > 0)[0]
-        return self.data, reward, done, {"safe": safe, "reach": reach,
-                                         "collision": collision_agent}
+        safe = 1.0 - collision.sum() / self.num_agents
+        return self.data, reward.detach(), done, {
+            "safe": safe, "reach": reach, "collision": collision}
 
     def forward_graph(self, data: GraphBatch, action: Tensor) -> GraphBatch:
         # reference gcbf/env/simple_car.py:178-194

@@ -162,11 +162,26 @@ class SimpleDrone(MultiAgentEnv):
         return self._data
 
     # ------------------------------------------------------------------ step
-    def step(self, action: Tensor) -> Tuple[GraphBatch, np.ndarray, bool, dict]:
+    def _get_K_tensor(self) -> Tensor:
+        if self._K is None:
+            self.u_ref(self._data)
+        return self._K.contiguous()
+
+    _finish_fused_step = SimpleCar._finish_fused_step
+
+    def step(self, action: Tensor) -> Tuple[GraphBatch, Tensor, bool, dict]:
         self._t += 1
+        # fused single-kernel path on GPU (ops/hip/env_step.hip)
+        out = self._ops.env_step_fused(
+            "drone", self._data.states, self._goal, action,
+            self._get_K_tensor(), self.dt, self._params["drone_radius"],
+            self._params["speed_limit"], self._params["dist2goal"], 10.0)
+        if out is not None:
+            return self._finish_fused_step(out)
+
         # reference gcbf/env/simple_drone.py:191-234
         reward_action = -torch.norm(action, dim=1) * 0.001
-        action = action + self.u_ref(self._data)
+        action = action + self._step_u_ref()
         lower_lim, upper_lim = self.action_lim
         action = torch.clamp(action, lower_lim, upper_lim)
         am = self._data.agent_mask
@@ -192,10 +207,9 @@ class SimpleDrone(MultiAgentEnv):
         reward_reach = (reach.int() - prev_reach.int()) * 10
         reward = reward_reach + reward_collision + reward_step + reward_action
 
-        safe = float(1.0 - collision.sum() / self.num_agents)
-        collision_agent = torch.where(collision > 0)[0]
-        return self.data, reward.detach().cpu().numpy(), done, {
-            "safe": safe, "reach": reach, "collision": collision_agent}
+        safe = 1.0 - collision.sum() / self.num_agents
+        return self.data, reward.detach(), done, {
+            "safe": safe, "reach": reach, "collision": collision}
 
     def forward_graph(self, data: GraphBatch, action: Tensor) -> GraphBatch:
         action = action + self.u_ref(data)
@@ -220,6 +234,7 @@ class SimpleDrone(MultiAgentEnv):
     add_communication_links_batched = SimpleCar.add_communication_links
     _fused_mask = SimpleCar._fused_mask
     _mask_rows = SimpleCar._mask_rows
+    _step_u_ref = SimpleCar._step_u_ref
 
     @property
     def state_lim(self) -> Tuple[Tensor, Tensor]:

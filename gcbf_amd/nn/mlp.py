@@ -62,23 +62,28 @@ class SNLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         W = self.weight_orig
-        if self.training:
-            with torch.no_grad():
-                # fp32 power iteration regardless of autocast
-                Wf = W.detach().float()
-                u, v = self.weight_u, self.weight_v
-                for _ in range(self.n_power_iterations):
-                    v = F.normalize(Wf.t().mv(u), dim=0, eps=self.eps)
-                    u = F.normalize(Wf.mv(v), dim=0, eps=self.eps)
-                self.weight_u.copy_(u)
-                self.weight_v.copy_(v)
-        # σ differentiable through W (u, v constants) — same as torch's hook.
-        # clone() so later in-place buffer updates (next forward's power
-        # iteration) don't invalidate this graph's saved tensors.
-        u = self.weight_u.clone().to(W.dtype)
-        v = self.weight_v.clone().to(W.dtype)
-        sigma = torch.dot(u, torch.mv(W, v))
-        return F.linear(x, W / sigma, self.bias)
+        # everything except the big GEMM runs with autocast OFF: the matvecs
+        # are tiny in fp32 but autocast would route them to a pathologically
+        # slow bf16 GEMV (~20x the whole layer's cost, measured on MI355X)
+        with torch.autocast(W.device.type if W.device.type != "cpu"
+                            else "cpu", enabled=False):
+            if self.training:
+                with torch.no_grad():
+                    Wf = W.detach().float()
+                    u, v = self.weight_u, self.weight_v
+                    for _ in range(self.n_power_iterations):
+                        v = F.normalize(Wf.t().mv(u), dim=0, eps=self.eps)
+                        u = F.normalize(Wf.mv(v), dim=0, eps=self.eps)
+                    self.weight_u.copy_(u)
+                    self.weight_v.copy_(v)
+            # σ differentiable through W (u, v constants) — same as torch's
+            # hook.  clone() so the next forward's in-place buffer update
+            # doesn't invalidate this graph's saved tensors.
+            u = self.weight_u.clone()
+            v = self.weight_v.clone()
+            sigma = torch.dot(u, torch.mv(W.float(), v))
+            W_eff = W / sigma
+        return F.linear(x, W_eff, self.bias)
 
     def extra_repr(self) -> str:
         return f"in_features={self.in_features}, " \

@@ -216,6 +216,23 @@ def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],
                                       diag_offset)
 
 
+def env_step_fused(kind: str, *args):
+    """Fused single-graph rollout step (env_step.hip); None on CPU/no-ext.
+
+    kind: "dubins" | "car" | "drone"; args are forwarded to the binding.
+    Returns (new_states, u_ref_next, reward, reach, collision).
+    """
+    states = args[0]
+    if not states.is_cuda:
+        return None
+    ext = _require_ext(f"{kind}_step")
+    if ext is None:
+        return None
+    fn = {"dubins": ext.dubins_step, "car": ext.car_step,
+          "drone": ext.drone_step}[kind]
+    return fn(*[a.contiguous() if torch.is_tensor(a) else a for a in args])
+
+
 # env kinds understood by the fused mask kernel (mirrors masks.hip)
 ENV_CAR = 0
 ENV_DUBINS = 1

@@ -26,6 +26,25 @@ extern "C" void launch_fused_masks(const float* states, bool* safe,
                                    bool* unsafe, bool* collision, int B,
                                    int N, int n_rec, int S, float r, int kind,
                                    hipStream_t stream);
+extern "C" void launch_dubins_step(const float* states, const float* goal,
+                                   const float* action, float* new_states,
+                                   float* u_ref_next, float* reward,
+                                   bool* reach, bool* collision, int N, int n,
+                                   float dt, float r, float sl, float d2g,
+                                   float act_lim, hipStream_t stream);
+extern "C" void launch_car_step(const float* states, const float* goal,
+                                const float* action, const float* K,
+                                float* new_states, float* u_ref_next,
+                                float* reward, bool* reach, bool* collision,
+                                int N, float dt, float r, float sl, float d2g,
+                                float act_lim, hipStream_t stream);
+extern "C" void launch_drone_step(const float* states, const float* goal,
+                                  const float* action, const float* K,
+                                  float* new_states, float* u_ref_next,
+                                  float* reward, bool* reach,
+                                  bool* collision, int N, int n, float dt,
+                                  float r, float sl, float d2g,
+                                  float act_lim, hipStream_t stream);
 
 namespace {
 
@@ -129,11 +148,78 @@ std::vector<torch::Tensor> fused_masks(torch::Tensor states, int64_t B,
     return {t_safe, t_uns, t_coll};
 }
 
+struct StepOut {
+    torch::Tensor new_states, u_ref_next, reward, reach, collision;
+};
+
+static std::vector<torch::Tensor> alloc_step_out(torch::Tensor states,
+                                                 int64_t n, int64_t A) {
+    auto f = states.options();
+    auto b = states.options().dtype(torch::kBool);
+    return {torch::empty_like(states), torch::empty({n, A}, f),
+            torch::empty({n}, f), torch::empty({n}, b),
+            torch::empty({n}, b)};
+}
+
+std::vector<torch::Tensor> dubins_step(torch::Tensor states,
+                                       torch::Tensor goal,
+                                       torch::Tensor action, double dt,
+                                       double r, double sl, double d2g,
+                                       double act_lim) {
+    CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action);
+    const int64_t N = states.size(0), n = action.size(0);
+    auto out = alloc_step_out(states, n, 2);
+    launch_dubins_step(states.data_ptr<float>(), goal.data_ptr<float>(),
+                       action.data_ptr<float>(), out[0].data_ptr<float>(),
+                       out[1].data_ptr<float>(), out[2].data_ptr<float>(),
+                       out[3].data_ptr<bool>(), out[4].data_ptr<bool>(),
+                       (int)N, (int)n, (float)dt, (float)r, (float)sl,
+                       (float)d2g, (float)act_lim, current_stream());
+    return out;
+}
+
+std::vector<torch::Tensor> car_step(torch::Tensor states, torch::Tensor goal,
+                                    torch::Tensor action, torch::Tensor K,
+                                    double dt, double r, double sl,
+                                    double d2g, double act_lim) {
+    CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action); CHECK_IN(K);
+    const int64_t N = states.size(0);
+    auto out = alloc_step_out(states, N, 2);
+    launch_car_step(states.data_ptr<float>(), goal.data_ptr<float>(),
+                    action.data_ptr<float>(), K.data_ptr<float>(),
+                    out[0].data_ptr<float>(), out[1].data_ptr<float>(),
+                    out[2].data_ptr<float>(), out[3].data_ptr<bool>(),
+                    out[4].data_ptr<bool>(), (int)N, (float)dt, (float)r,
+                    (float)sl, (float)d2g, (float)act_lim, current_stream());
+    return out;
+}
+
+std::vector<torch::Tensor> drone_step(torch::Tensor states,
+                                      torch::Tensor goal,
+                                      torch::Tensor action, torch::Tensor K,
+                                      double dt, double r, double sl,
+                                      double d2g, double act_lim) {
+    CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action); CHECK_IN(K);
+    const int64_t N = states.size(0), n = action.size(0);
+    auto out = alloc_step_out(states, n, 3);
+    launch_drone_step(states.data_ptr<float>(), goal.data_ptr<float>(),
+                      action.data_ptr<float>(), K.data_ptr<float>(),
+                      out[0].data_ptr<float>(), out[1].data_ptr<float>(),
+                      out[2].data_ptr<float>(), out[3].data_ptr<bool>(),
+                      out[4].data_ptr<bool>(), (int)N, (int)n, (float)dt,
+                      (float)r, (float)sl, (float)d2g, (float)act_lim,
+                      current_stream());
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_masks", &fused_masks,
           "batched safe/unsafe/collision agent masks in one pass");
+    m.def("dubins_step", &dubins_step, "fused DubinsCar rollout step");
+    m.def("car_step", &car_step, "fused SimpleCar rollout step");
+    m.def("drone_step", &drone_step, "fused SimpleDrone rollout step");
     m.def("segment_attn_fwd", &segment_attn_fwd,
           "fused scatter-softmax + weighted scatter-sum (forward)");
     m.def("segment_attn_bwd", &segment_attn_bwd,

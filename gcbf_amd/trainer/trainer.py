@@ -53,7 +53,8 @@ class Trainer:
 
         verbose = None
         for step in range(start_step, steps + 1):
-            data.update(u_ref=self.env.u_ref(data))
+            if data.u_ref is None:
+                data.update(u_ref=self.env.u_ref(data))
             action = self.algo.step(data, prob=1 - (step - 1) / steps)
             next_data, reward, done, info = self.env.step(action)
             next_data.update(u_ref=self.env.u_ref(next_data))
@@ -129,7 +130,7 @@ class Trainer:
                 data.update(u_ref=self.env_test.u_ref(data))
                 action = self.algo.apply(data)
                 data, reward, done, info = self.env_test.step(action)
-                epi_reward += np.mean(reward)
+                epi_reward += float(reward.float().mean())
                 if "collision" in info:
                     safe_agent[info["collision"].cpu()] = False
                 if "reach" in info:

@@ -95,7 +95,7 @@ def eval_ctrl_epi(controller: Callable, env: MultiAgentEnv, seed: int = 0,
             states.append(data.states.unsqueeze(0).cpu())
         next_data, reward, done, info = env.step(action)
         epi_length += 1
-        epi_reward += np.mean(reward)
+        epi_reward += float(reward.float().mean())
         if "collision" in info:
             safe_agent[info["collision"].cpu()] = False
             safe = torch.ones(env.num_agents).bool()

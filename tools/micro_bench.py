@@ -4,7 +4,11 @@ Run on an MI355X box:  python tools/micro_bench.py
 Prints per-op timings to locate where rollout/update time goes and whether
 bf16 GEMMs are healthy on this torch build.
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
@@ -148,8 +152,35 @@ def rollout_breakdown():
           f" ms")
 
 
+def shape_churn():
+    """Is bf16 slow when the GEMM M dimension changes every call
+    (hipBLASLt algorithm search per novel shape)?"""
+    print("== shape churn: K=2048 N=2048, M varies per call ==")
+    w32 = torch.randn(2048, 2048, device="cuda")
+    w16 = w32.bfloat16()
+    for dt, w in [("fp32", w32), ("bf16", w16)]:
+        xs = [torch.randn(9000 + 37 * i, 2048, device="cuda",
+                          dtype=w.dtype) for i in range(40)]
+        # first pass: novel shapes every call
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for x in xs:
+            F.linear(x, w)
+        torch.cuda.synchronize()
+        t_novel = (time.perf_counter() - t0) / len(xs) * 1e3
+        # second pass: same shapes again (cached)
+        t0 = time.perf_counter()
+        for x in xs:
+            F.linear(x, w)
+        torch.cuda.synchronize()
+        t_cached = (time.perf_counter() - t0) / len(xs) * 1e3
+        print(f"  {dt}: novel shapes {t_novel:7.3f} ms/call, "
+              f"repeat shapes {t_cached:7.3f} ms/call")
+
+
 if __name__ == "__main__":
     assert torch.cuda.is_available()
+    shape_churn()
     gemm_suite()
     relu_cost()
     sn_cost()
