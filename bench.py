@@ -72,13 +72,27 @@ def run(args):
     data = env.reset()
     prof = {"rollout_s": 0.0, "update_s": 0.0, "updates": 0}
 
+    engine = None
+    if use_cuda and not args.no_capture:
+        from gcbf_amd.rollout import RolloutEngine, engine_supported
+        if engine_supported(env, algo):
+            engine = RolloutEngine(env, algo)
+            if rank == 0:
+                print("# hipGraph-captured rollout engine active",
+                      flush=True)
+
     def one_step(step, timed=False):
         nonlocal data
         t0 = time.perf_counter() if timed else 0.0
-        data.update(u_ref=env.u_ref(data))
-        action = algo.step(data, prob=1 - (step - 1) / total_schedule)
-        next_data, reward, done, info = env.step(action)
-        data = env.reset() if done else next_data
+        if engine is not None:
+            done = engine.step(prob=1 - (step - 1) / total_schedule)
+            if done:
+                engine.reload()
+        else:
+            data.update(u_ref=env.u_ref(data))
+            action = algo.step(data, prob=1 - (step - 1) / total_schedule)
+            next_data, reward, done, info = env.step(action)
+            data = env.reset() if done else next_data
         if algo.is_update(step):
             if timed and use_cuda:
                 torch.cuda.synchronize(device)
@@ -161,4 +175,6 @@ if __name__ == "__main__":
                    choices=[None, "bf16", "fp32"])
     p.add_argument("--profile", action="store_true", default=False,
                    help="print rollout/update time split to stderr")
+    p.add_argument("--no-capture", action="store_true", default=False,
+                   help="disable the hipGraph rollout engine")
     run(p.parse_args())

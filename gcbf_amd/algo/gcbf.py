@@ -59,7 +59,7 @@ class CBFGNN(nn.Module):
     def forward(self, data: GraphBatch) -> Tensor:
         x = self.feat_transformer.module_0(
             data.x, data.edge_attr, data.edge_index,
-            node_mask=data.agent_mask)
+            node_mask=data.agent_mask, seg_dst=data.seg_dst)
         return self.feat_2_CBF(x)
 
     def attention(self, data: GraphBatch) -> Tensor:

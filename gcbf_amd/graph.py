@@ -43,7 +43,7 @@ class GraphBatch:
     """
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
-                 "u_ref", "ptr", "_dst_ptr")
+                 "u_ref", "ptr", "_dst_ptr", "seg_dst")
 
     def __init__(
             self,
@@ -68,6 +68,10 @@ class GraphBatch:
                                device=states.device)
         self.ptr = ptr
         self._dst_ptr = None  # lazy CSR pointer over destinations
+        # optional: segment destinations for aggregation when edge buffers
+        # are padded to a fixed capacity (pad entries carry a sentinel id
+        # past the last node); None means edge_index[1] is used directly
+        self.seg_dst = None
 
     # ------------------------------------------------------------------ sizes
     @property

@@ -73,10 +73,12 @@ class CBFGNNLayer(nn.Module):
                         hidden_layers=(128, 128), limit_lip=False))
 
     def forward(self, x: Tensor, edge_attr: Tensor, edge_index: Tensor,
-                node_mask: Optional[Tensor] = None) -> Tensor:
+                node_mask: Optional[Tensor] = None,
+                seg_dst: Optional[Tensor] = None) -> Tensor:
         num_nodes = x.shape[0]
         msg = self.phi(_gather_edge_inputs(x, edge_attr, edge_index))
-        aggr = self.aggr_module(msg, edge_index[1], num_nodes)
+        dst = edge_index[1] if seg_dst is None else seg_dst
+        aggr = self.aggr_module(msg, dst, num_nodes)
         gamma_in = torch.cat([aggr, x], dim=1)
         if node_mask is not None:
             gamma_in = gamma_in[node_mask]
@@ -107,10 +109,12 @@ class ControllerGNNLayer(nn.Module):
                         hidden_layers=(128, 128)))
 
     def forward(self, x: Tensor, edge_attr: Tensor, edge_index: Tensor,
-                node_mask: Optional[Tensor] = None) -> Tensor:
+                node_mask: Optional[Tensor] = None,
+                seg_dst: Optional[Tensor] = None) -> Tensor:
         num_nodes = x.shape[0]
         msg = self.phi(_gather_edge_inputs(x, edge_attr, edge_index))
-        aggr = self.aggr_module(msg, edge_index[1], num_nodes)
+        dst = edge_index[1] if seg_dst is None else seg_dst
+        aggr = self.aggr_module(msg, dst, num_nodes)
         gamma_in = torch.cat([aggr, x], dim=1)
         if node_mask is not None:
             gamma_in = gamma_in[node_mask]

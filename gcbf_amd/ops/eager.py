@@ -29,23 +29,26 @@ def segment_softmax(gate: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
     by the per-segment max; empty segments produce no outputs).
     """
     gate = gate.squeeze(-1)
-    seg_max = torch.full((num_nodes,), float("-inf"),
+    # one extra row absorbs sentinel destinations (padded edge lists use
+    # dst == num_nodes for invalid entries)
+    seg_max = torch.full((num_nodes + 1,), float("-inf"),
                          dtype=gate.dtype, device=gate.device)
     seg_max = seg_max.scatter_reduce(0, dst, gate.detach(), reduce="amax",
                                      include_self=True)
     shifted = gate - seg_max.index_select(0, dst)
     ex = shifted.exp()
-    denom = torch.zeros(num_nodes, dtype=gate.dtype, device=gate.device)
+    denom = torch.zeros(num_nodes + 1, dtype=gate.dtype, device=gate.device)
     denom = denom.index_add(0, dst, ex)
     att = ex / (denom.index_select(0, dst) + 1e-16)
     return att.unsqueeze(-1)
 
 
 def segment_sum(values: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
-    """Sum of (E, D) edge values into (num_nodes, D) per destination."""
-    out = torch.zeros(num_nodes, values.shape[1],
+    """Sum of (E, D) edge values into (num_nodes, D) per destination.
+    A sentinel row absorbs dst == num_nodes (padded edge lists)."""
+    out = torch.zeros(num_nodes + 1, values.shape[1],
                       dtype=values.dtype, device=values.device)
-    return out.index_add(0, dst, values)
+    return out.index_add(0, dst, values)[:num_nodes]
 
 
 def segment_attn_aggregate(msg: Tensor, gate: Tensor, dst: Tensor,

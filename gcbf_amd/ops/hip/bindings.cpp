@@ -22,6 +22,11 @@ extern "C" void launch_radius_fill(const float* pos, const float* states,
                                    int N, int n_rec, int P, int S, int A,
                                    float r, int topk, int attr_kind,
                                    hipStream_t stream);
+extern "C" void launch_pad_edges(const int* offsets, const int* counts,
+                                 long* edge_index, long* seg,
+                                 float* edge_attr, int* e_count, int rows,
+                                 long E_max, long N_total, int A,
+                                 hipStream_t stream);
 extern "C" void launch_fused_masks(const float* states, bool* safe,
                                    bool* unsafe, bool* collision, int B,
                                    int N, int n_rec, int S, float r, int kind,
@@ -148,6 +153,45 @@ std::vector<torch::Tensor> fused_masks(torch::Tensor states, int64_t B,
     return {t_safe, t_uns, t_coll};
 }
 
+std::vector<torch::Tensor> build_graph_padded(torch::Tensor pos,
+                                              torch::Tensor states,
+                                              int64_t B, int64_t n_rec,
+                                              double r, int64_t topk,
+                                              int64_t attr_kind,
+                                              int64_t attr_dim,
+                                              int64_t E_max) {
+    // capture-safe variant: fixed E_max edge buffers, no host sync.
+    CHECK_IN(pos);
+    CHECK_IN(states);
+    const int64_t N = pos.size(0) / B;
+    const int64_t P = pos.size(1), S = states.size(1);
+    auto stream = current_stream();
+    auto counts = torch::empty({B * n_rec},
+                               pos.options().dtype(torch::kInt32));
+    launch_radius_count(pos.data_ptr<float>(), counts.data_ptr<int>(),
+                        (int)B, (int)N, (int)n_rec, (int)P, (float)r,
+                        (int)topk, stream);
+    auto incl = counts.cumsum(0, torch::kInt32);
+    auto offsets = (incl - counts).contiguous();
+    auto edge_index = torch::empty({2, E_max},
+                                   pos.options().dtype(torch::kInt64));
+    auto seg = torch::empty({E_max}, pos.options().dtype(torch::kInt64));
+    auto edge_attr = torch::empty({E_max, attr_dim}, pos.options());
+    auto e_count = torch::empty({1}, pos.options().dtype(torch::kInt32));
+    launch_radius_fill(pos.data_ptr<float>(), states.data_ptr<float>(),
+                       offsets.data_ptr<int>(),
+                       edge_index.data_ptr<int64_t>(),
+                       edge_attr.data_ptr<float>(), E_max, (int)B, (int)N,
+                       (int)n_rec, (int)P, (int)S, (int)attr_dim, (float)r,
+                       (int)topk, (int)attr_kind, stream);
+    launch_pad_edges(offsets.data_ptr<int>(), counts.data_ptr<int>(),
+                     edge_index.data_ptr<int64_t>(),
+                     seg.data_ptr<int64_t>(), edge_attr.data_ptr<float>(),
+                     e_count.data_ptr<int>(), (int)(B * n_rec), E_max,
+                     B * N, (int)attr_dim, stream);
+    return {edge_index, seg, edge_attr, e_count};
+}
+
 struct StepOut {
     torch::Tensor new_states, u_ref_next, reward, reach, collision;
 };
@@ -226,4 +270,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused scatter-softmax + weighted scatter-sum (backward)");
     m.def("build_graph", &build_graph,
           "batched dense radius graph + edge_attr (count/scan/fill)");
+    m.def("build_graph_padded", &build_graph_padded,
+          "capture-safe radius graph into fixed E_max buffers");
 }

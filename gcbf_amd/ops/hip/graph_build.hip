@@ -159,6 +159,45 @@ extern "C" __global__ void radius_fill(
     }
 }
 
+// Pad a built edge list out to a fixed capacity so every consumer shape is
+// static (hipGraph capture of the rollout loop).  Real edges [0, E) get
+// seg = dst (for CSR segment ops); pad entries get src = dst = 0 (a valid
+// gather index) and seg = N_total (a sentinel past the last node, so CSR
+// segments skip them), edge_attr = 0.  Also publishes E to a device scalar.
+extern "C" __global__ void pad_edges(
+        const int* __restrict__ offsets,   // (rows,) exclusive scan
+        const int* __restrict__ counts,    // (rows,)
+        long* __restrict__ edge_index,     // (2, E_max)
+        long* __restrict__ seg,            // (E_max,)
+        float* __restrict__ edge_attr,     // (E_max, A)
+        int* __restrict__ e_count,         // (1,) out
+        int rows, long E_max, long N_total, int A) {
+    const long E = (long)offsets[rows - 1] + counts[rows - 1];
+    const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx == 0) *e_count = (int)E;
+    if (idx >= E_max) return;
+    if (idx < E) {
+        seg[idx] = edge_index[E_max + idx];  // dst
+    } else {
+        edge_index[idx] = 0;
+        edge_index[E_max + idx] = 0;
+        seg[idx] = N_total;
+        for (int c = 0; c < A; ++c) edge_attr[idx * A + c] = 0.f;
+    }
+}
+
+extern "C" void launch_pad_edges(const int* offsets, const int* counts,
+                                 long* edge_index, long* seg,
+                                 float* edge_attr, int* e_count, int rows,
+                                 long E_max, long N_total, int A,
+                                 hipStream_t stream) {
+    const int threads = 256;
+    const int blocks = (int)((E_max + threads - 1) / threads);
+    hipLaunchKernelGGL(pad_edges, dim3(blocks > 0 ? blocks : 1),
+                       dim3(threads), 0, stream, offsets, counts, edge_index,
+                       seg, edge_attr, e_count, rows, E_max, N_total, A);
+}
+
 // ------------------------------------------------------------- launchers
 extern "C" void launch_radius_count(const float* pos, int* counts, int B,
                                     int N, int n_rec, int P, float r,

@@ -1,0 +1,201 @@
+"""hipGraph-captured rollout engine.
+
+The rollout loop is launch/Python-bound on GPU (~25 kernels for the actor
+GNN + graph rebuild + env step per env step).  This engine captures ONE
+whole training rollout step into a hipGraph (torch.cuda.CUDAGraph == hipGraph
+on ROCm) and replays it per step:
+
+    actor forward (over padded, fixed-capacity edge buffers)
+    × exploration gate (host-written device scalar)
+    → fused env-step kernel (dynamics + rewards + next u_ref)
+    → capture-safe padded graph rebuild (fixed E_max buffers)
+    → flag publication ([edge_count, reach_all, unsafe_any] in one int32
+      tensor; the host reads it once per step)
+
+Static shapes come from padding: edge buffers hold E_max = n·(N−1) entries,
+pad entries carry src=dst=0 (valid gather index) and a segment sentinel
+id == num_nodes so the CSR attention aggregation skips them; the GEMMs run
+over E_max rows always (tiny at rollout sizes).  Exploration, episode
+termination and replay-buffer appends stay on the host.
+
+Weights are read by address, so Adam steps between replays take effect
+without re-capture.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from . import ops
+from .graph import GraphBatch
+
+
+def engine_supported(env, algo) -> bool:
+    from .algo.gcbf import GCBF
+    from .algo.macbf import MACBF
+    if not torch.cuda.is_available() or not ops.hip_available():
+        return False
+    if not isinstance(algo, GCBF) or isinstance(algo, MACBF):
+        return False
+    return env._mode == "train" and env._max_neighbors is None
+
+
+class RolloutEngine:
+
+    def __init__(self, env, algo):
+        self.env = env
+        self.algo = algo
+        self.device = env.device
+        self.n = env.num_agents
+        self.pos_dim = 3 if env.state_dim == 6 else 2
+        self.edge_dim = env.edge_dim
+
+        data = env.data if env.data is not None else env.reset()
+        self.N = data.num_nodes
+        self.E_max = self.n * (self.N - 1)
+
+        from gcbf_amd import _C
+        self._ext = _C
+
+        kind_by_class = {"SimpleCar": "car", "DubinsCar": "dubins",
+                         "SimpleDrone": "drone"}
+        self.kind = kind_by_class[type(env).__name__]
+
+        dev = self.device
+        self.states = data.states.clone().contiguous()
+        self.x = data.x.clone()
+        self.agent_mask = (None if data.agent_mask is None
+                           else data.agent_mask.clone())
+        self.goal = env._goal.clone().contiguous()
+        self.u_ref = env.u_ref(data).clone().contiguous()
+        self.ei = torch.zeros(2, self.E_max, dtype=torch.long, device=dev)
+        self.seg = torch.full((self.E_max,), self.N, dtype=torch.long,
+                              device=dev)
+        self.ea = torch.zeros(self.E_max, self.edge_dim, device=dev)
+        self.explore = torch.ones(1, device=dev)
+        self.flags = torch.zeros(3, dtype=torch.int32, device=dev)
+        self.E = 0
+
+        if self.kind in ("car", "drone"):
+            self.K = env._get_K_tensor().clone().contiguous()
+
+        self._load_graph_from_env()
+        self._capture()
+
+    # ----------------------------------------------------------------- body
+    def _step_args(self, action):
+        p = self.env.params
+        r_key = "drone_radius" if self.kind == "drone" else "car_radius"
+        act_lim = {"car": 10.0, "dubins": 2.0, "drone": 10.0}[self.kind]
+        base = [self.states, self.goal, action]
+        if self.kind in ("car", "drone"):
+            base.append(self.K)
+        return base + [self.env.dt, p[r_key], p["speed_limit"],
+                       p["dist2goal"], act_lim]
+
+    def _body(self):
+        data = GraphBatch(x=self.x, pos=self.states[:, :self.pos_dim],
+                          states=self.states, edge_index=self.ei,
+                          edge_attr=self.ea, agent_mask=self.agent_mask,
+                          u_ref=self.u_ref)
+        data.seg_dst = self.seg
+        with torch.no_grad():
+            action = self.algo.actor(data) * self.explore
+            unsafe_any = self.env.unsafe_mask(data).any()
+            out = ops.env_step_fused(self.kind, *self._step_args(action))
+            new_states, u_ref_next, reward, reach, collision = out
+            ei, seg, ea, ecount = self._ext.build_graph_padded(
+                new_states[:, :self.pos_dim].contiguous(), new_states, 1,
+                self.n if self.agent_mask is not None else self.N,
+                self.env.params["comm_radius"], -1,
+                self.env._attr_kind, self.edge_dim, self.E_max)
+            # publish flags, then advance the static buffers
+            self.flags[0].copy_(ecount[0])
+            self.flags[1].copy_(reach.all().to(torch.int32))
+            self.flags[2].copy_(unsafe_any.to(torch.int32))
+            self.states.copy_(new_states)
+            self.u_ref.copy_(u_ref_next)
+            self.ei.copy_(ei)
+            self.seg.copy_(seg)
+            self.ea.copy_(ea)
+
+    def _capture(self):
+        saved = (self.states.clone(), self.u_ref.clone(), self.ei.clone(),
+                 self.seg.clone(), self.ea.clone())
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._body()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._body()
+        # restore pre-warmup state
+        self.states.copy_(saved[0])
+        self.u_ref.copy_(saved[1])
+        self.ei.copy_(saved[2])
+        self.seg.copy_(saved[3])
+        self.ea.copy_(saved[4])
+
+    # ---------------------------------------------------------- env plumbing
+    def _load_graph_from_env(self):
+        """Refresh static buffers from the env's current graph (after
+        reset)."""
+        data = self.env.data
+        self.states.copy_(data.states)
+        self.goal.copy_(self.env._goal)
+        self.u_ref.copy_(self.env.u_ref(data))
+        ei, seg, ea, ecount = self._ext.build_graph_padded(
+            data.pos.contiguous(), data.states.contiguous(), 1,
+            self.n if self.agent_mask is not None else self.N,
+            self.env.params["comm_radius"], -1,
+            self.env._attr_kind, self.edge_dim, self.E_max)
+        self.ei.copy_(ei)
+        self.seg.copy_(seg)
+        self.ea.copy_(ea)
+        self.E = int(ecount.item())
+
+    def reload(self):
+        self.env.reset()
+        self._load_graph_from_env()
+
+    # ----------------------------------------------------------------- step
+    def step(self, prob: float) -> bool:
+        """One training env step.  Returns done."""
+        # clone the CURRENT graph for the replay buffer before the replay
+        # overwrites the static buffers
+        E = self.E
+        snap = GraphBatch(
+            x=self.x,  # static content, shared
+            pos=self.states[:, :self.pos_dim].clone(),
+            states=self.states.clone(),
+            edge_index=self.ei[:, :E].clone(),
+            edge_attr=self.ea[:E].clone(),
+            agent_mask=self.agent_mask,
+            u_ref=self.u_ref.clone())
+
+        self.explore.fill_(0.0 if np.random.rand() < prob else 1.0)
+        self.graph.replay()
+
+        flags = self.flags.cpu()  # ONE host sync per step
+        self.E = int(flags[0])
+        reach_all = bool(flags[1])
+        is_safe = not bool(flags[2])
+
+        self.algo.buffer.append(snap, is_safe)
+
+        self.env._t += 1
+        done = self.env._t >= self.env.max_episode_steps or reach_all
+        if not done:
+            # keep the env object's view of the world consistent (cheap:
+            # shares the engine's buffers; env methods are not used for
+            # captured stepping)
+            self.env._data = GraphBatch(
+                x=self.x, pos=self.states[:, :self.pos_dim],
+                states=self.states, edge_index=self.ei, edge_attr=self.ea,
+                agent_mask=self.agent_mask, u_ref=self.u_ref)
+            self.env._data.seg_dst = self.seg
+        return done

@@ -44,22 +44,44 @@ class Trainer:
             self.model_dir = os.path.join(log_dir, "models")
             self.writer = NullWriter()
 
+    def _make_engine(self):
+        """hipGraph-captured rollout when supported (GPU + HIP ext + GCBF)."""
+        try:
+            from ..rollout import RolloutEngine, engine_supported
+            if engine_supported(self.env, self.algo):
+                if self.env.data is None:
+                    self.env.reset()
+                return RolloutEngine(self.env, self.algo)
+        except Exception as e:
+            if self.rank == 0:
+                print(f"> rollout capture unavailable ({e}); "
+                      f"using the eager loop", flush=True)
+        return None
+
     def train(self, steps: int, eval_interval: int, eval_epi: int,
-              start_step: int = 1):
+              start_step: int = 1, capture: bool = True):
         start_time = time.time()
         data = self.env.reset()
+        engine = self._make_engine() if capture else None
+        if engine is not None and self.rank == 0:
+            print("> hipGraph-captured rollout engine active", flush=True)
         last_report = start_time
         steps_since_report = 0
 
         verbose = None
         for step in range(start_step, steps + 1):
-            if data.u_ref is None:
-                data.update(u_ref=self.env.u_ref(data))
-            action = self.algo.step(data, prob=1 - (step - 1) / steps)
-            next_data, reward, done, info = self.env.step(action)
-            next_data.update(u_ref=self.env.u_ref(next_data))
-            self.algo.post_step(data, action, reward, done, next_data)
-            data = self.env.reset() if done else next_data
+            if engine is not None:
+                done = engine.step(prob=1 - (step - 1) / steps)
+                if done:
+                    engine.reload()
+            else:
+                if data.u_ref is None:
+                    data.update(u_ref=self.env.u_ref(data))
+                action = self.algo.step(data, prob=1 - (step - 1) / steps)
+                next_data, reward, done, info = self.env.step(action)
+                next_data.update(u_ref=self.env.u_ref(next_data))
+                self.algo.post_step(data, action, reward, done, next_data)
+                data = self.env.reset() if done else next_data
 
             if self.algo.is_update(step):
                 verbose = self.algo.update(step, self.writer)

@@ -275,8 +275,9 @@ def test_fused_env_step_matches_python(env_name, n, obs):
     act = torch.randn(n, env.action_dim, device=dev) * 0.1
 
     def snapshot():
+        obs = getattr(env, "_obs", None)
         return (env._data.replace(), env._t,
-                None if env._obs is None else env._obs.clone())
+                None if obs is None else obs.clone())
 
     def restore(s):
         env._data, env._t, obs_s = s
@@ -325,3 +326,82 @@ def test_fused_env_step_full_episode():
         if done:
             break
     assert done
+
+
+def test_rollout_engine_matches_eager_loop():
+    """Captured rollout vs. the eager loop: same seeds, same exploration
+    stream -> trajectories must agree closely for many steps."""
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.rollout import RolloutEngine, engine_supported
+    from gcbf_amd.trainer.utils import set_seed
+    import numpy as np
+    dev = torch.device("cuda")
+
+    def build():
+        set_seed(3)
+        env = make_env("DubinsCar", 16, dev)
+        env.train()
+        algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                         env.action_dim, dev, batch_size=512)
+        return env, algo
+
+    # eager loop
+    env_a, algo_a = build()
+    data = env_a.reset()
+    np.random.seed(7)
+    states_a = []
+    for step in range(40):
+        if data.u_ref is None:
+            data.update(u_ref=env_a.u_ref(data))
+        a = algo_a.step(data, prob=0.5)
+        data, r, done, info = env_a.step(a)
+        states_a.append(data.states.clone())
+        if done:
+            break
+
+    # captured loop (same weights by construction: same seed -> same init)
+    env_b, algo_b = build()
+    env_b.reset()
+    assert engine_supported(env_b, algo_b)
+    np.random.seed(7)
+    eng = RolloutEngine(env_b, algo_b)
+    states_b = []
+    for step in range(len(states_a)):
+        done = eng.step(prob=0.5)
+        states_b.append(eng.states.clone())
+        if done:
+            break
+
+    assert len(states_a) == len(states_b)
+    for t, (sa, sb) in enumerate(zip(states_a, states_b)):
+        assert torch.allclose(sa, sb, atol=5e-4), \
+            (t, (sa - sb).abs().max())
+    # buffers got the same number of graphs with matching edge counts
+    assert algo_a.buffer.size == algo_b.buffer.size
+    for ga, gb in zip(algo_a.buffer.data, algo_b.buffer.data):
+        assert ga.num_edges == gb.num_edges
+
+
+def test_rollout_engine_update_interleave():
+    """Engine + updates: weights change between replays and the captured
+    graph must pick them up (actor output changes)."""
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.rollout import RolloutEngine
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(0)
+    dev = torch.device("cuda")
+    env = make_env("DubinsCar", 16, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=32)
+    env.reset()
+    eng = RolloutEngine(env, algo)
+    for step in range(1, 65):
+        done = eng.step(prob=0.1)
+        if done:
+            eng.reload()
+        if algo.is_update(step):
+            out = algo.update(step, None)
+    assert all(0 <= v <= 1 for v in out.values())
