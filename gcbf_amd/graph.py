@@ -43,7 +43,7 @@ class GraphBatch:
     """
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
-                 "u_ref", "ptr", "_dst_ptr", "seg_dst")
+                 "u_ref", "_ptr", "_dst_ptr", "seg_dst")
 
     def __init__(
             self,
@@ -63,10 +63,9 @@ class GraphBatch:
         self.edge_attr = edge_attr
         self.agent_mask = agent_mask
         self.u_ref = u_ref
-        if ptr is None:
-            ptr = torch.tensor([0, states.shape[0]], dtype=torch.long,
-                               device=states.device)
-        self.ptr = ptr
+        # ptr materializes lazily: a host->device copy here would break
+        # hipGraph capture (single graphs never need the tensor)
+        self._ptr = ptr
         self._dst_ptr = None  # lazy CSR pointer over destinations
         # optional: segment destinations for aggregation when edge buffers
         # are padded to a fixed capacity (pad entries carry a sentinel id
@@ -74,6 +73,18 @@ class GraphBatch:
         self.seg_dst = None
 
     # ------------------------------------------------------------------ sizes
+    @property
+    def ptr(self) -> Tensor:
+        if self._ptr is None:
+            self._ptr = torch.tensor([0, self.states.shape[0]],
+                                     dtype=torch.long,
+                                     device=self.states.device)
+        return self._ptr
+
+    @ptr.setter
+    def ptr(self, value):
+        self._ptr = value
+
     @property
     def num_nodes(self) -> int:
         return self.states.shape[0]
@@ -84,7 +95,7 @@ class GraphBatch:
 
     @property
     def num_graphs(self) -> int:
-        return self.ptr.shape[0] - 1
+        return 1 if self._ptr is None else self._ptr.shape[0] - 1
 
     @property
     def nodes_per_graph(self) -> int:
@@ -114,7 +125,7 @@ class GraphBatch:
             edge_attr=kwargs.get("edge_attr", self.edge_attr),
             agent_mask=kwargs.get("agent_mask", self.agent_mask),
             u_ref=kwargs.get("u_ref", self.u_ref),
-            ptr=kwargs.get("ptr", self.ptr),
+            ptr=kwargs.get("ptr", self._ptr),
         )
         return out
 
@@ -215,11 +226,12 @@ class GraphBatch:
             None if self.edge_attr is None else self.edge_attr.detach(),
             self.agent_mask,
             None if self.u_ref is None else self.u_ref.detach(),
-            self.ptr)
+            self._ptr)
 
     def to(self, device) -> "GraphBatch":
         def mv(t):
             return None if t is None else t.to(device)
         return GraphBatch(mv(self.x), mv(self.pos), mv(self.states),
                           mv(self.edge_index), mv(self.edge_attr),
-                          mv(self.agent_mask), mv(self.u_ref), mv(self.ptr))
+                          mv(self.agent_mask), mv(self.u_ref),
+                          mv(self._ptr))

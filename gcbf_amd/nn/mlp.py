@@ -60,11 +60,15 @@ class SNLinear(nn.Module):
                               torch.mv(self.weight_orig, self.weight_v))
         return self.weight_orig / sigma
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def effective_weight(self) -> torch.Tensor:
+        """W/σ with gradient through W (u, v constants), running one power
+        iteration first in training mode — torch's spectral_norm semantics.
+
+        Runs with autocast OFF: the matvecs are tiny in fp32 but autocast
+        would route them to a pathologically slow bf16 GEMV (~20x the whole
+        layer's cost, measured on MI355X).
+        """
         W = self.weight_orig
-        # everything except the big GEMM runs with autocast OFF: the matvecs
-        # are tiny in fp32 but autocast would route them to a pathologically
-        # slow bf16 GEMV (~20x the whole layer's cost, measured on MI355X)
         with torch.autocast(W.device.type if W.device.type != "cpu"
                             else "cpu", enabled=False):
             if self.training:
@@ -76,14 +80,15 @@ class SNLinear(nn.Module):
                         u = F.normalize(Wf.mv(v), dim=0, eps=self.eps)
                     self.weight_u.copy_(u)
                     self.weight_v.copy_(v)
-            # σ differentiable through W (u, v constants) — same as torch's
-            # hook.  clone() so the next forward's in-place buffer update
-            # doesn't invalidate this graph's saved tensors.
+            # clone() so the next forward's in-place buffer update doesn't
+            # invalidate this graph's saved tensors.
             u = self.weight_u.clone()
             v = self.weight_v.clone()
             sigma = torch.dot(u, torch.mv(W.float(), v))
-            W_eff = W / sigma
-        return F.linear(x, W_eff, self.bias)
+            return W / sigma
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.linear(x, self.effective_weight(), self.bias)
 
     def extra_repr(self) -> str:
         return f"in_features={self.in_features}, " \
@@ -124,6 +129,24 @@ class MLP(nn.Module):
         if output_activation is not None:
             layers.append(output_activation)
         self.net = nn.Sequential(*layers)
+        # execution plan for the fused MFMA path: (linear, fused act code)
+        from .fused import ACT_NONE, ACT_RELU, ACT_TANH
+        plan = []
+        mods = list(self.net)
+        i = 0
+        while i < len(mods):
+            m = mods[i]
+            if isinstance(m, (nn.Linear, SNLinear)):
+                act = ACT_NONE
+                if i + 1 < len(mods) and isinstance(mods[i + 1], nn.ReLU):
+                    act = ACT_RELU
+                    i += 1
+                elif i + 1 < len(mods) and isinstance(mods[i + 1], nn.Tanh):
+                    act = ACT_TANH
+                    i += 1
+                plan.append((m, act))
+            i += 1
+        self._plan = plan
 
     # GEMM row-bucketing: graph batches make M (edge/node counts) unique on
     # almost every call, and hipBLASLt pays a per-novel-shape algorithm
@@ -131,8 +154,14 @@ class MLP(nn.Module):
     # cached).  Padding M up to a bucket multiple makes shapes recur; the
     # zero rows are sliced off after the chain (values/grads unchanged).
     BUCKET = 256
+    # set True (by utils.amp.enable_bf16) to route big layers through the
+    # hand-written MFMA kernel with fused bias+activation epilogues
+    fused_mfma = False
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fused_mfma and x.is_cuda:
+            from . import fused
+            return fused.run_plan(self._plan, x, bucket=self.BUCKET)
         M = x.shape[0]
         if x.is_cuda and M > 0 and M % self.BUCKET != 0:
             pad = self.BUCKET - M % self.BUCKET

@@ -22,6 +22,11 @@ extern "C" void launch_radius_fill(const float* pos, const float* states,
                                    int N, int n_rec, int P, int S, int A,
                                    float r, int topk, int attr_kind,
                                    hipStream_t stream);
+extern "C" void launch_fused_linear_bf16(const void* A, const void* W,
+                                         const float* bias, void* Cb,
+                                         float* Cf, int M, int N, int K,
+                                         int act, int out_f32,
+                                         hipStream_t stream);
 extern "C" void launch_pad_edges(const int* offsets, const int* counts,
                                  long* edge_index, long* seg,
                                  float* edge_attr, int* e_count, int rows,
@@ -192,6 +197,36 @@ std::vector<torch::Tensor> build_graph_padded(torch::Tensor pos,
     return {edge_index, seg, edge_attr, e_count};
 }
 
+torch::Tensor fused_linear(torch::Tensor x, torch::Tensor w,
+                           c10::optional<torch::Tensor> bias, int64_t act,
+                           bool out_f32) {
+    CHECK_IN(x);
+    CHECK_IN(w);
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+    TORCH_CHECK(w.scalar_type() == torch::kBFloat16, "w must be bf16");
+    const int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+    TORCH_CHECK(w.size(1) == K, "K mismatch");
+    TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+                "fused_linear needs M%128==0, N%128==0, K%64==0 (got ", M,
+                "x", K, " -> ", N, ")");
+    const float* bptr = nullptr;
+    if (bias.has_value()) {
+        CHECK_IN(bias.value());
+        TORCH_CHECK(bias->scalar_type() == torch::kFloat32,
+                    "bias must be fp32");
+        bptr = bias->data_ptr<float>();
+    }
+    auto out = torch::empty({M, N}, x.options().dtype(
+        out_f32 ? torch::kFloat32 : torch::kBFloat16));
+    launch_fused_linear_bf16(
+        x.data_ptr(), w.data_ptr(), bptr,
+        out_f32 ? nullptr : out.data_ptr(),
+        out_f32 ? out.data_ptr<float>() : nullptr,
+        (int)M, (int)N, (int)K, (int)act, out_f32 ? 1 : 0,
+        current_stream());
+    return out;
+}
+
 struct StepOut {
     torch::Tensor new_states, u_ref_next, reward, reach, collision;
 };
@@ -272,4 +307,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "batched dense radius graph + edge_attr (count/scan/fill)");
     m.def("build_graph_padded", &build_graph_padded,
           "capture-safe radius graph into fixed E_max buffers");
+    m.def("fused_linear", &fused_linear,
+          "MFMA bf16 GEMM with fused bias + activation epilogue");
 }

@@ -25,10 +25,24 @@ def _wrap_forward(module: torch.nn.Module):
     module.forward = wrapped
 
 
-def enable_bf16(algo):
-    """Enable bf16 autocast on the algorithm's networks (GPU only)."""
+def enable_bf16(algo, fused_mfma=None):
+    """Enable bf16 compute on the algorithm's networks (GPU only):
+    autocast at the module boundary, plus the hand-written MFMA fused
+    linear kernels for the big MLP layers (GCBF_AMD_FUSED=0 disables)."""
+    import os
+
+    from .. import ops
+    from ..nn.mlp import MLP
+
     if not torch.cuda.is_available():
         return algo
     for m in (algo.cbf, algo.actor):
         _wrap_forward(m)
+    if fused_mfma is None:
+        fused_mfma = os.environ.get("GCBF_AMD_FUSED", "1") == "1"
+    if fused_mfma and ops.hip_available():
+        for mod in (algo.cbf, algo.actor):
+            for m in mod.modules():
+                if isinstance(m, MLP):
+                    m.fused_mfma = True
     return algo
