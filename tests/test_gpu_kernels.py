@@ -405,3 +405,89 @@ def test_rollout_engine_update_interleave():
         if algo.is_update(step):
             out = algo.update(step, None)
     assert all(0 <= v <= 1 for v in out.values())
+
+
+class TestFusedLinear:
+    """MFMA bf16 GEMM kernel (ops/hip/fused_linear.hip) numerics."""
+
+    def test_identity_layout(self):
+        # A = I picks out rows of W^T: catches any fragment-layout mixup
+        from gcbf_amd import _C
+        M, K, N = 128, 64, 128
+        A = torch.zeros(M, K, device="cuda")
+        for i in range(K):
+            A[i, i] = 1.0
+        W = torch.randn(N, K, device="cuda")  # asymmetric
+        out = _C.fused_linear(A.bfloat16(), W.bfloat16(), None, 0, True)
+        ref = W.t().float()
+        assert torch.allclose(out[:K], W.bfloat16().float().t(), atol=1e-3), \
+            (out[:K] - ref[:K]).abs().max()
+        assert out[K:].abs().max() == 0
+
+    @pytest.mark.parametrize("M,K,N", [(256, 64, 128), (512, 2048, 2048),
+                                       (1024, 2048, 256), (256, 320, 2048),
+                                       (2048, 256, 128)])
+    def test_matches_blas_bf16(self, M, K, N):
+        from gcbf_amd import _C
+        torch.manual_seed(M + K + N)
+        A = torch.randn(M, K, device="cuda")
+        W = torch.randn(N, K, device="cuda") / (K ** 0.5)
+        b = torch.randn(N, device="cuda")
+        out = _C.fused_linear(A.bfloat16(), W.bfloat16(), b, 0, True)
+        ref = (A.bfloat16() @ W.bfloat16().t()).float() + b
+        err = (out - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err < 2e-2 * max(scale, 1.0), (err, scale)
+
+    def test_relu_tanh_epilogues(self):
+        from gcbf_amd import _C
+        torch.manual_seed(0)
+        A = torch.randn(256, 128, device="cuda")
+        W = torch.randn(128, 128, device="cuda") / 12.0
+        b = torch.randn(128, device="cuda")
+        ref = (A.bfloat16() @ W.bfloat16().t()).float() + b
+        out_r = _C.fused_linear(A.bfloat16(), W.bfloat16(), b, 1, True)
+        assert torch.allclose(out_r, torch.relu(ref), atol=2e-2)
+        out_t = _C.fused_linear(A.bfloat16(), W.bfloat16(), b, 2, True)
+        assert torch.allclose(out_t, torch.tanh(ref), atol=2e-2)
+
+    def test_bf16_output(self):
+        from gcbf_amd import _C
+        A = torch.randn(128, 64, device="cuda")
+        W = torch.randn(128, 64, device="cuda")
+        out = _C.fused_linear(A.bfloat16(), W.bfloat16(), None, 0, False)
+        assert out.dtype == torch.bfloat16
+
+    def test_mlp_fused_plan_matches_eager(self):
+        from gcbf_amd.nn import MLP
+        torch.manual_seed(0)
+        mlp = MLP(13, 256, (2048, 2048), limit_lip=True).cuda()
+        mlp.eval()
+        x = torch.randn(700, 13, device="cuda")
+        with torch.no_grad():
+            ref = mlp(x)            # eager fp32
+            mlp.fused_mfma = True
+            out = mlp(x)            # MFMA bf16 path
+        rel = (out - ref).abs().max() / ref.abs().max().clamp_min(1e-3)
+        assert rel < 0.05, rel
+
+    def test_fused_backward_grads(self):
+        from gcbf_amd.nn.fused import fused_linear_act
+        torch.manual_seed(0)
+        M, K, N = 256, 128, 128
+        x = torch.randn(M, K, device="cuda", requires_grad=True)
+        w = torch.randn(N, K, device="cuda", requires_grad=True) / 10
+        b = torch.randn(N, device="cuda", requires_grad=True)
+        out = fused_linear_act(x, w, b, 1, True)
+        g = torch.randn_like(out)
+        out.backward(g)
+
+        x2 = x.detach().bfloat16().float().requires_grad_(True)
+        w2 = w.detach().bfloat16().float().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        ref = torch.relu(torch.nn.functional.linear(x2, w2, b2))
+        ref.backward(g)
+        assert torch.allclose(x.grad, x2.grad, atol=0.05,
+                              rtol=0.05), (x.grad - x2.grad).abs().max()
+        assert torch.allclose(w.grad, w2.grad, atol=0.05, rtol=0.05)
+        assert torch.allclose(b.grad, b2.grad, atol=0.05, rtol=0.05)
