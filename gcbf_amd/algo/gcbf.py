@@ -282,15 +282,20 @@ class GCBF(Algorithm):
         ok = (max_val_h_dot[:, 0] <= 0)
         action = torch.where(ok.unsqueeze(1), nominal, action)
 
-        actions: List[Tensor] = list(torch.split(action, 1, dim=0))
-        for a in actions:
-            a.requires_grad_(True)
-        optim = [Adam((a,), lr=lr) for a in actions]
+        # Vectorized per-agent Adam (equivalent to the reference's list of
+        # per-agent optimizers, gcbf/algo/gcbf.py:275-307: Adam state and
+        # step counts advance only for violating agents; gradients keep
+        # accumulating for agents that are not stepped).
+        action = action.clone().requires_grad_(True)
+        exp_avg = torch.zeros_like(action)
+        exp_avg_sq = torch.zeros_like(action)
+        step_cnt = torch.zeros(action.shape[0], 1, device=action.device)
+        grad_buf = torch.zeros_like(action)
+        b1, b2, adam_eps = 0.9, 0.999, 1e-8
 
         i_iter = 0
         max_iter = 30
         while True:
-            action = torch.cat(actions, dim=0)
             data_next = self._env.forward_graph(data, action)
             h_next = self.cbf(data_next)
             h_dot = (h_next - h) / self._env.dt
@@ -298,15 +303,23 @@ class GCBF(Algorithm):
             loss_h_dot = torch.mean(max_val_h_dot)
             if loss_h_dot <= 0 or i_iter > max_iter:
                 break
-            val_agent = torch.nonzero(max_val_h_dot)[:, 0]
-            for i in val_agent:
-                optim[i].zero_grad(set_to_none=True)
-            loss_h_dot.backward()
-            for i in val_agent:
-                optim[i].step()
-                with torch.no_grad():
-                    grad = actions[i].grad
-                    actions[i] -= rand * lr * torch.randn_like(grad) * grad
+            val = (max_val_h_dot[:, 0] > 0).unsqueeze(1)
+            g, = torch.autograd.grad(loss_h_dot, action)
+            with torch.no_grad():
+                grad_buf = torch.where(val, g, grad_buf + g)
+                step_cnt = step_cnt + val
+                exp_avg = torch.where(
+                    val, b1 * exp_avg + (1 - b1) * grad_buf, exp_avg)
+                exp_avg_sq = torch.where(
+                    val, b2 * exp_avg_sq + (1 - b2) * grad_buf ** 2,
+                    exp_avg_sq)
+                bc1 = 1 - b1 ** step_cnt.clamp(min=1)
+                bc2 = 1 - b2 ** step_cnt.clamp(min=1)
+                upd = (lr / bc1) * exp_avg / (
+                    (exp_avg_sq / bc2).sqrt() + adam_eps)
+                noise = rand * lr * torch.randn_like(grad_buf) * grad_buf
+                action -= torch.where(val, upd + noise,
+                                      torch.zeros_like(upd))
             i_iter += 1
 
         return action.detach()
