@@ -76,10 +76,16 @@ def run(args):
     if use_cuda and not args.no_capture:
         from gcbf_amd.rollout import RolloutEngine, engine_supported
         if engine_supported(env, algo):
-            engine = RolloutEngine(env, algo)
-            if rank == 0:
-                print("# hipGraph-captured rollout engine active",
-                      flush=True)
+            try:
+                engine = RolloutEngine(env, algo)
+                if rank == 0:
+                    print("# hipGraph-captured rollout engine active",
+                          flush=True)
+            except Exception as e:
+                engine = None
+                if rank == 0:
+                    print(f"# rollout capture unavailable ({e}); "
+                          f"eager loop", flush=True)
 
     def one_step(step, timed=False):
         nonlocal data

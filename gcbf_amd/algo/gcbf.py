@@ -57,9 +57,12 @@ class CBFGNN(nn.Module):
                               output_activation=nn.Tanh())
 
     def forward(self, data: GraphBatch) -> Tensor:
+        nm = data.agent_mask
+        if data.agents_first_n is not None:
+            nm = data.agents_first_n
         x = self.feat_transformer.module_0(
             data.x, data.edge_attr, data.edge_index,
-            node_mask=data.agent_mask, seg_dst=data.seg_dst)
+            node_mask=nm, seg_dst=data.seg_dst)
         return self.feat_2_CBF(x)
 
     def attention(self, data: GraphBatch) -> Tensor:

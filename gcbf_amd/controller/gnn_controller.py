@@ -36,7 +36,10 @@ class GNNController(MultiAgentController):
                                  hidden_layers=(512, 128, 32))
 
     def forward(self, data: GraphBatch) -> Tensor:
+        nm = data.agent_mask
+        if data.agents_first_n is not None:
+            nm = data.agents_first_n
         x = self.feat_transformer.module_0(
             data.x, data.edge_attr, data.edge_index,
-            node_mask=data.agent_mask, seg_dst=data.seg_dst)
+            node_mask=nm, seg_dst=data.seg_dst)
         return self.feat_2_action(torch.cat([x, data.u_ref], dim=1))

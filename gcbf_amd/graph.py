@@ -43,7 +43,7 @@ class GraphBatch:
     """
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
-                 "u_ref", "_ptr", "_dst_ptr", "seg_dst")
+                 "u_ref", "_ptr", "_dst_ptr", "seg_dst", "agents_first_n")
 
     def __init__(
             self,
@@ -71,6 +71,10 @@ class GraphBatch:
         # are padded to a fixed capacity (pad entries carry a sentinel id
         # past the last node); None means edge_index[1] is used directly
         self.seg_dst = None
+        # optional: number of leading agent rows (single graphs lay agents
+        # first) — lets models use a static slice instead of boolean-mask
+        # indexing, which is data-dependent and not hipGraph-capturable
+        self.agents_first_n = None
 
     # ------------------------------------------------------------------ sizes
     @property

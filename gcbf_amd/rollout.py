@@ -108,6 +108,7 @@ class RolloutEngine:
                           edge_attr=self.ea, agent_mask=self.agent_mask,
                           u_ref=self.u_ref)
         data.seg_dst = self.seg
+        data.agents_first_n = self.n if self.agent_mask is not None else None
         with torch.no_grad():
             action = self.algo.actor(data) * self.explore
             unsafe_any = self.env.unsafe_mask(data).any()
