@@ -180,8 +180,10 @@ class DubinsCar(SimpleCar):
         agent_mask = torch.zeros(self.num_agents + n_obs, dtype=torch.bool,
                                  device=self.device)
         agent_mask[:self.num_agents] = True
-        return GraphBatch(x=x, pos=states[:, :2], states=states,
-                          agent_mask=agent_mask)
+        g = GraphBatch(x=x, pos=states[:, :2], states=states,
+                       agent_mask=agent_mask)
+        g.agents_first_n = self.num_agents
+        return g
 
     # ------------------------------------------------------------------ step
     def step(self, action: Tensor) -> Tuple[GraphBatch, Tensor, bool, dict]:
@@ -208,6 +210,7 @@ class DubinsCar(SimpleCar):
 
         data = GraphBatch(
             x=self._data.x, pos=state[:, :2], states=state, agent_mask=am)
+        data.agents_first_n = self.num_agents
         self._obs = state[~am]
         self._data = self.add_communication_links(data)
 

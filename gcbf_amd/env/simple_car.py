@@ -145,6 +145,8 @@ class SimpleCar(MultiAgentEnv):
         data = GraphBatch(
             x=self._data.x, pos=new_states[:, :pos_dim], states=new_states,
             agent_mask=self._data.agent_mask)
+        if data.agent_mask is not None:
+            data.agents_first_n = self.num_agents
         self._data = self.add_communication_links(data)
         self._data.u_ref = u_ref_next
         done = bool(self._t >= self.max_episode_steps or reach.all())

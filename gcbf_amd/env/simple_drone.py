@@ -158,6 +158,7 @@ class SimpleDrone(MultiAgentEnv):
             pos=torch.cat([states[:, :3], self._obs[:, :3]], dim=0),
             states=torch.cat([states, self._obs], dim=0),
             agent_mask=agent_mask)
+        data.agents_first_n = self.num_agents
         self._data = self.add_communication_links(data)
         return self._data
 
@@ -193,6 +194,7 @@ class SimpleDrone(MultiAgentEnv):
 
         data = GraphBatch(x=self._data.x, pos=state[:, :3], states=state,
                           agent_mask=am)
+        data.agents_first_n = self.num_agents
         self._data = self.add_communication_links(data)
 
         time_up = self._t >= self.max_episode_steps
