@@ -73,8 +73,17 @@ def _pad_k(x: Tensor, w: Tensor):
     return F.pad(x, (0, pad)), F.pad(w, (0, pad))
 
 
+import os
+
+# below this many rows the 128x128-tile kernel cannot fill 256 CUs and
+# hipBLASLt's small-M kernels win (measured: rollout at M=256 is ~35% slower
+# on the tile kernel; update batches at M>=8K are faster on it)
+FUSED_MIN_M = int(os.environ.get("GCBF_AMD_FUSED_MIN_M", "1024"))
+
+
 def shapes_ok(m_padded: int, n: int) -> bool:
-    return m_padded % 128 == 0 and n % 128 == 0 and n >= 128
+    return (m_padded % 128 == 0 and n % 128 == 0 and n >= 128
+            and m_padded >= FUSED_MIN_M)
 
 
 def run_plan(plan, x: Tensor, bucket: int = 256) -> Tensor:

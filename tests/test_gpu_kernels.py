@@ -482,15 +482,17 @@ class TestFusedLinear:
         g = torch.randn_like(out)
         out.backward(g)
 
-        x2 = x.detach().bfloat16().float().requires_grad_(True)
-        w2 = w.detach().bfloat16().float().requires_grad_(True)
-        b2 = b.detach().clone().requires_grad_(True)
-        ref = torch.relu(torch.nn.functional.linear(x2, w2, b2))
-        ref.backward(g)
-        # bf16 matmul grads vs fp32 reference: tolerance scales with the
-        # reduction length (M=256 for dw)
-        assert torch.allclose(x.grad, x2.grad, atol=0.3,
-                              rtol=0.05), (x.grad - x2.grad).abs().max()
-        assert torch.allclose(w.grad, w2.grad, atol=0.5, rtol=0.05), \
-            (w.grad - w2.grad).abs().max()
-        assert torch.allclose(b.grad, b2.grad, atol=0.3, rtol=0.05)
+        # reference gradients USING THE SAME relu mask as the fused output
+        # (bf16 rounding flips marginal pre-activations, so masks from an
+        # fp32 recomputation differ on a few elements — that is inherent to
+        # bf16 training, not a kernel bug)
+        gm = g * (out > 0)
+        gm16 = gm.bfloat16()
+        dx_ref = (gm16 @ w.detach().bfloat16()).float()
+        dw_ref = (gm16.t() @ x.detach().bfloat16()).float()
+        db_ref = gm.sum(0)
+        assert torch.allclose(x.grad, dx_ref, atol=1e-3), \
+            (x.grad - dx_ref).abs().max()
+        assert torch.allclose(w.grad, dw_ref, atol=1e-3), \
+            (w.grad - dw_ref).abs().max()
+        assert torch.allclose(b.grad, db_ref, atol=1e-3)
