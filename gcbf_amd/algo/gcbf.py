@@ -133,7 +133,18 @@ class GCBF(Algorithm):
         acc_safe = acc_unsafe = acc_h_dot = torch.zeros((), dtype=torch.float)
         logs = []  # deferred scalars, synced once at the end
 
+        prof = None
+        if os.environ.get("GCBF_AMD_UPDATE_PROF") == "1":
+            import time as _time
+            prof = {"sample": 0.0, "batch": 0.0, "fwd": 0.0, "mask": 0.0,
+                    "hdot": 0.0, "bwd": 0.0, "opt": 0.0}
+
+            def _tick():
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize()
+                return _time.perf_counter()
         for i_inner in range(inner_iter):
+            t0 = _tick() if prof else 0
             # sample segments from the current buffer and the replay memory
             if self.memory.size == 0:
                 graph_list = self.buffer.sample(self.batch_size // 5, seg_len)
@@ -142,12 +153,18 @@ class GCBF(Algorithm):
                 prev = self.memory.sample(
                     self.batch_size // 5 - self.batch_size // 10, seg_len, True)
                 graph_list = curr + prev
+            if prof:
+                t1 = _tick(); prof["sample"] += t1 - t0; t0 = t1
 
             graphs = GraphBatch.from_list(graph_list)
             graphs.edge_attr.requires_grad_(True)
+            if prof:
+                t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1
             with trace_range("gcbf/forward"):
                 h = self.cbf(graphs)
                 actions = self.actor(graphs)
+            if prof:
+                t1 = _tick(); prof["fwd"] += t1 - t0; t0 = t1
 
             # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177)
             unsafe_mask = self._env.unsafe_mask(graphs)
@@ -171,6 +188,8 @@ class GCBF(Algorithm):
                 loss_safe = torch.tensor(0.0).type_as(h)
                 acc_safe = torch.tensor(1.0).type_as(h)
 
+            if prof:
+                t1 = _tick(); prof["mask"] += t1 - t0; t0 = t1
             # ḣ condition with the re-link residue trick
             # (reference gcbf/algo/gcbf.py:191-209): the VALUE reflects the
             # re-linked next graph, the GRADIENT flows through the
@@ -188,6 +207,8 @@ class GCBF(Algorithm):
             residue = (h_dot_new_link - h_dot).detach()
             h_dot = residue + h_dot
 
+            if prof:
+                t1 = _tick(); prof["hdot"] += t1 - t0; t0 = t1
             loss_h_dot = torch.mean(torch.relu(-h_dot - alpha * h + eps))
             acc_h_dot = torch.mean(
                 torch.greater_equal(h_dot + alpha * h, 0).type_as(h_dot))
@@ -206,6 +227,8 @@ class GCBF(Algorithm):
             if self.grad_sync is not None:
                 with trace_range("gcbf/grad_allreduce"):
                     self.grad_sync()
+            if prof:
+                t1 = _tick(); prof["bwd"] += t1 - t0; t0 = t1
             with trace_range("gcbf/optim"):
                 torch.nn.utils.clip_grad_norm_(self.cbf.parameters(), 1e-3)
                 torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
@@ -216,6 +239,13 @@ class GCBF(Algorithm):
                 loss_unsafe.detach(), loss_safe.detach(),
                 loss_h_dot.detach(), loss_action.detach(),
                 acc_unsafe.detach(), acc_safe.detach(), acc_h_dot.detach()]))
+
+        if prof:
+            t1 = _tick(); prof["opt"] += t1 - t0
+            import sys
+            print("# update prof: " + " ".join(
+                f"{k}={v * 1000:.1f}ms" for k, v in prof.items()),
+                file=sys.stderr, flush=True)
 
         # one host sync for the whole update's scalars
         log_vals = torch.stack(logs).cpu()
