@@ -147,6 +147,12 @@ class MACBF(GCBF):
                 for k, name in enumerate(names):
                     writer.add_scalar(name, float(log_vals[i_inner, k]), t)
 
+        # refresh bf16 weight mirrors (captured rollout graphs read them
+        # by address; Python is skipped during replay)
+        from ..nn.fused import sync_bf16_mirrors
+        sync_bf16_mirrors(self.actor)
+        sync_bf16_mirrors(self.cbf)
+
         self.memory.merge(self.buffer)
         self.buffer.clear()
         return {
