@@ -476,7 +476,7 @@ class TestFusedLinear:
         torch.manual_seed(0)
         M, K, N = 256, 128, 128
         x = torch.randn(M, K, device="cuda", requires_grad=True)
-        w = torch.randn(N, K, device="cuda", requires_grad=True) / 10
+        w = (torch.randn(N, K, device="cuda") / 10).requires_grad_(True)
         b = torch.randn(N, device="cuda", requires_grad=True)
         out = fused_linear_act(x, w, b, 1, True)
         g = torch.randn_like(out)
