@@ -63,8 +63,10 @@ def _mirror(lin):
                        (0, K64 - K)).contiguous()
         bb_new = None if lin.bias is None else \
             lin.bias.detach().bfloat16().contiguous()
+        # copy=True: .float() on an fp32 bias would alias the master and
+        # the refresh copy_ would mutate it (version-bump cascade)
         b32_new = None if lin.bias is None else \
-            lin.bias.detach().float().contiguous()
+            lin.bias.detach().to(torch.float32, copy=True).contiguous()
         if cache is not None:
             # reuse storage so captured graphs pick up the refresh
             cache[1].copy_(wb_new)
@@ -102,7 +104,7 @@ class _FusedLinearAct(torch.autograd.Function):
         xb = x if x.dtype == torch.bfloat16 else x.bfloat16()
         xb = _pad_k_x(xb, wb.shape[1]).contiguous()
         if bias is not None and b32 is None:
-            b32 = bias.detach().float().contiguous()
+            b32 = bias.detach().to(torch.float32, copy=True).contiguous()
         out = _ext().fused_linear(xb, wb, b32, act, out_fp32)
         ctx.save_for_backward(xb, wb, out)
         ctx.act = act
