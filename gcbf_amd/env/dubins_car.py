@@ -76,8 +76,10 @@ class DubinsCar(SimpleCar):
         return 5
 
     # -------------------------------------------------------------- dynamics
-    def dynamics(self, data: GraphBatch, u: Tensor) -> Tensor:
+    def dynamics(self, data: GraphBatch, u) -> Tensor:
         # reference gcbf/env/dubins_car.py:110-132
+        if not torch.is_tensor(u):
+            raise NotImplementedError  # same as the reference (:112)
         agent_mask = data.agent_mask
         s = data.states
         sl = self._params["speed_limit"]

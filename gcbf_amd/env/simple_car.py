@@ -66,7 +66,17 @@ class SimpleCar(MultiAgentEnv):
         }
 
     # -------------------------------------------------------------- dynamics
-    def dynamics(self, data: GraphBatch, u: Tensor) -> Tensor:
+    def dynamics(self, data: GraphBatch, u) -> Tensor:
+        if not torch.is_tensor(u):
+            # symbolic/numpy overload for external CBF-QP use (reference
+            # gcbf/env/simple_car.py:80-87 with a cvxpy Expression): works
+            # with any object supporting "@" (cvxpy Expression, numpy array)
+            x = data.states.cpu().detach().numpy()
+            A = np.zeros((self.state_dim, self.state_dim))
+            A[0, 2] = 1.0
+            A[1, 3] = 1.0
+            B = np.array([[1, 0], [0, 1], [0, 0], [0, 0]])
+            return x @ A.T + u @ B.T
         x = data.states
         return torch.cat([x[:, 2:], u], dim=1)
 

@@ -93,8 +93,13 @@ class SimpleDrone(MultiAgentEnv):
         return B
 
     # -------------------------------------------------------------- dynamics
-    def dynamics(self, data: GraphBatch, u: Tensor) -> Tensor:
+    def dynamics(self, data: GraphBatch, u) -> Tensor:
         # reference gcbf/env/simple_drone.py:103-120
+        if not torch.is_tensor(u):
+            # symbolic/numpy overload (cvxpy Expression or numpy array)
+            A = self._A.cpu().numpy()
+            B = self._B.cpu().numpy()
+            return data.states.cpu().detach().numpy() @ A.T + u @ B.T
         am = data.agent_mask
         s = data.states
         xdot = s @ self._A.t()

@@ -133,3 +133,69 @@ def rejection_sample_positions(n: int, dim: int, side_length: float,
         pts[i] = cand
         i += 1
     return pts
+
+
+def create_point_cloud_surface(vertices: Tensor, r: float) -> Tensor:
+    """Point cloud covering a quadrilateral surface (reference
+    gcbf/env/utils.py:119-131)."""
+    points = []
+    length = torch.norm(vertices[:, 1, :] - vertices[:, 0, :])
+    width = torch.norm(vertices[:, 2, :] - vertices[:, 1, :])
+    for i in range(1, int(length // (2 * r))):
+        for j in range(int(width // (2 * r) + 1)):
+            points.append(
+                vertices[:, 0, :]
+                + i * 2 * r * (vertices[:, 1, :] - vertices[:, 0, :]) / length
+                + j * 2 * r * (vertices[:, 2, :] - vertices[:, 1, :]) / width)
+    for vertex in vertices:
+        for i in range(4):
+            points.append(vertex[i, :].unsqueeze(0))
+    return torch.cat(points, dim=0)
+
+
+def create_point_cloud(vertices: Tensor, r: float, dim: int = 2) -> Tensor:
+    """Sample obstacle boundary points every 2r (reference
+    gcbf/env/utils.py:134-150)."""
+    points = []
+    if dim == 2:
+        for i in range(vertices.shape[0]):
+            points.append(vertices[i, :])
+            j = i + 1 if i < vertices.shape[0] - 1 else 0
+            direction = (vertices[j, :] - vertices[i, :]) / torch.norm(
+                vertices[j, :] - vertices[i, :])
+            while torch.norm(points[-1] - vertices[j, :]) > 2 * r:
+                points.append(points[-1] + 2 * r * direction)
+        points = torch.stack(points, dim=0)
+    elif dim == 3:
+        surface_nodes = [[0, 1, 2, 3], [4, 5, 6, 7], [0, 4, 5, 1],
+                         [1, 2, 6, 5], [2, 6, 7, 3], [0, 3, 7, 4]]
+        points = create_point_cloud_surface(vertices[surface_nodes, :], r)
+    else:
+        raise NotImplementedError
+    return points
+
+
+def create_rectangle(center: Tensor, length: float, width: float,
+                     theta: float) -> Tensor:
+    """Rotated rectangle vertices (reference gcbf/env/utils.py:153-161)."""
+    vertices = torch.tensor([[length / 2, width / 2],
+                             [length / 2, -width / 2],
+                             [-length / 2, -width / 2],
+                             [-length / 2, width / 2]]).type_as(center)
+    rot = torch.tensor([[np.cos(theta), -np.sin(theta)],
+                        [np.sin(theta), np.cos(theta)]]).type_as(center)
+    return center + vertices @ rot
+
+
+def create_cuboid(center: Tensor, length: float, width: float, height: float,
+                  theta: float) -> Tensor:
+    """Rotated cuboid vertices (reference gcbf/env/utils.py:164-175)."""
+    signs = [(1, 1, 1), (1, -1, 1), (-1, -1, 1), (-1, 1, 1),
+             (1, 1, -1), (1, -1, -1), (-1, -1, -1), (-1, 1, -1)]
+    vertices = torch.tensor(
+        [[sx * length / 2, sy * width / 2, sz * height / 2]
+         for sx, sy, sz in signs]).type_as(center)
+    rot = torch.tensor([[np.cos(theta), -np.sin(theta), 0],
+                        [np.sin(theta), np.cos(theta), 0],
+                        [0, 0, 1]]).type_as(center)
+    return center + vertices @ rot

@@ -79,6 +79,13 @@ def train(args):
                      env.edge_dim, env.action_dim, device, args.batch_size,
                      hyperparams=hyper)
 
+    dtype = args.dtype or ("bf16" if use_cuda else "fp32")
+    if dtype == "bf16" and use_cuda and args.algo in ("gcbf", "macbf"):
+        from gcbf_amd.utils.amp import enable_bf16
+        enable_bf16(algo)
+        if rank == 0:
+            print("> bf16 compute enabled (MFMA fused linears + autocast)")
+
     if world_size > 1:
         broadcast_modules([algo.cbf, algo.actor])
         algo.grad_sync = GradSynchronizer([algo.cbf, algo.actor])
@@ -120,6 +127,8 @@ if __name__ == "__main__":
     parser.add_argument("--log-path", type=str, default="./logs")
     parser.add_argument("--batch-size", type=int, default=512)
     # additions over the reference
+    parser.add_argument("--dtype", type=str, default=None,
+                        choices=[None, "bf16", "fp32"])
     parser.add_argument("--resume", type=str, default=None,
                         help="path of a previous run's log dir to resume")
     parser.add_argument("--eval-epi", type=int, default=3)
