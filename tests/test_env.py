@@ -352,3 +352,12 @@ def test_relink_batched_matches_per_graph(env_name):
     ref = GraphBatch.from_list(parts)
     assert torch.equal(out.edge_index, ref.edge_index)
     assert torch.allclose(out.edge_attr, ref.edge_attr)
+
+
+def test_demo2_mode_limits_goal_distance():
+    env = _mk("SimpleCar", 4)
+    env.demo(2)
+    data = env.reset()
+    d = torch.norm(data.states[:, :2] - env._goal, dim=1)
+    assert (d <= env.params["max_distance"] * 1.5).all()
+    assert env.max_episode_steps == 2500

@@ -496,3 +496,61 @@ class TestFusedLinear:
         assert torch.allclose(w.grad, dw_ref, atol=1e-3), \
             (w.grad - dw_ref).abs().max()
         assert torch.allclose(b.grad, db_ref, atol=1e-3)
+
+
+@pytest.mark.parametrize("env_name,n,obs,algo_name,mn", [
+    ("SimpleCar", 16, None, "macbf", 12),     # BASELINE config 5
+    ("SimpleDrone", 16, None, "gcbf", None),  # config 3 (single-GPU slice)
+])
+def test_baseline_configs_train_gpu(env_name, n, obs, algo_name, mn):
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(0)
+    dev = torch.device("cuda")
+    env = make_env(env_name, n, dev, max_neighbors=mn)
+    env.train()
+    algo = make_algo(algo_name, env, n, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=32)
+    data = env.reset()
+    for step in range(1, 65):
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.6)
+        data, r, done, info = env.step(a)
+        if done:
+            data = env.reset()
+        if algo.is_update(step):
+            out = algo.update(step, None)
+    assert all(np.isfinite(v) for v in out.values())
+
+
+@pytest.mark.timeout(600)
+def test_stress_config_n256_updates():
+    """BASELINE config 4: DubinsCar n=256 + 32 obstacles — one update must
+    run within memory/time budget on one GPU."""
+    import numpy as np
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(0)
+    dev = torch.device("cuda")
+    e0 = make_env("DubinsCar", 256, dev)
+    p = e0.default_params
+    p["num_obs"] = 32
+    env = make_env("DubinsCar", 256, dev, params=p)
+    env.train()
+    algo = make_algo("gcbf", env, 256, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=32)
+    from gcbf_amd.utils.amp import enable_bf16
+    enable_bf16(algo)
+    data = env.reset()
+    assert data.num_nodes == 288
+    for step in range(1, 33):
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.9)
+        data, r, done, info = env.step(a)
+        if done:
+            data = env.reset()
+        if algo.is_update(step):
+            out = algo.update(step, None)
+    assert all(np.isfinite(v) for v in out.values())
