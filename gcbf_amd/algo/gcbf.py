@@ -161,8 +161,16 @@ class GCBF(Algorithm):
             if prof:
                 t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1
             with trace_range("gcbf/forward"):
-                h = self.cbf(graphs)
                 actions = self.actor(graphs)
+                # h and h_next in ONE doubled-batch CBF forward: halves the
+                # CBF forward/backward chains vs. the reference's separate
+                # calls (gcbf/algo/gcbf.py:161,194) and evaluates both sides
+                # of the finite difference under the SAME spectral-norm σ
+                graphs_next = self._env.forward_graph(graphs, actions)
+                both = GraphBatch.from_list([graphs, graphs_next])
+                h_both = self.cbf(both)
+                n_ag = h_both.shape[0] // 2
+                h, h_next = h_both[:n_ag], h_both[n_ag:]
             if prof:
                 t1 = _tick(); prof["fwd"] += t1 - t0; t0 = t1
 
@@ -195,12 +203,9 @@ class GCBF(Algorithm):
             # re-linked next graph, the GRADIENT flows through the
             # fixed-topology path.
             with trace_range("gcbf/h_dot"):
-                graphs_next = self._env.forward_graph(graphs, actions)
-                h_next = self.cbf(graphs_next)
                 with torch.no_grad():
                     relinked = self._env.add_communication_links_batched(
-                        self._env.forward_graph(graphs, actions.detach())
-                        .detach())
+                        graphs_next.detach())
                     h_next_new_link = self.cbf(relinked)
             h_dot = (h_next - h) / self._env.dt
             h_dot_new_link = (h_next_new_link - h) / self._env.dt
