@@ -14,14 +14,14 @@ import torch.multiprocessing as mp
 from gcbf_amd.parallel import GradSynchronizer, broadcast_modules
 
 
-def _init(rank, world):
+def _init(rank, world, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = os.environ.get("TEST_DDP_PORT", "29511")
+    os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
 
 
 def _worker_grad_sync(rank, world, q):
-    _init(rank, world)
+    _init(rank, world, 29511)
     torch.manual_seed(rank)  # different grads per rank
     m = torch.nn.Linear(4, 3)
     # identical weights
@@ -57,8 +57,7 @@ def test_grad_synchronizer_averages():
 
 
 def _worker_gcbf_dp(rank, world, q):
-    os.environ["TEST_DDP_PORT"] = "29513"
-    _init(rank, world)
+    _init(rank, world, 29519)
     from gcbf_amd.algo import make_algo
     from gcbf_amd.env import make_env
     from gcbf_amd.trainer.utils import set_seed
