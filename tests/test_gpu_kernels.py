@@ -4,6 +4,7 @@ All tests are @pytest.mark.gpu and run on an MI355X box; the ops dispatch
 layer routes CUDA tensors to gcbf_amd._C, so these exercise the native
 kernels end to end.
 """
+import numpy as np
 import pytest
 import torch
 
@@ -528,7 +529,6 @@ def test_baseline_configs_train_gpu(env_name, n, obs, algo_name, mn):
 def test_stress_config_n256_updates():
     """BASELINE config 4: DubinsCar n=256 + 32 obstacles — one update must
     run within memory/time budget on one GPU."""
-    import numpy as np
     from gcbf_amd.algo import make_algo
     from gcbf_amd.env import make_env
     from gcbf_amd.trainer.utils import set_seed
@@ -537,6 +537,10 @@ def test_stress_config_n256_updates():
     e0 = make_env("DubinsCar", 256, dev)
     p = e0.default_params
     p["num_obs"] = 32
+    # keep the n=16 packing density: 256 agents need a 16x16 area (the
+    # default 4x4 is near the random-sequential-packing jamming limit and
+    # the reference's rejection sampler would not terminate either)
+    p["area_size"] = 16.0
     env = make_env("DubinsCar", 256, dev, params=p)
     env.train()
     algo = make_algo("gcbf", env, 256, env.node_dim, env.edge_dim,
