@@ -81,7 +81,7 @@ class RolloutEngine:
         self.seg = torch.full((self.E_max,), self.N, dtype=torch.long,
                               device=dev)
         self.ea = torch.zeros(self.E_max, self.edge_dim, device=dev)
-        self.explore = torch.ones(1, device=dev)
+        self.zero_action = torch.zeros(self.n, env.action_dim, device=dev)
         self.flags = torch.zeros(3, dtype=torch.int32, device=dev)
         self.E = 0
 
@@ -102,7 +102,7 @@ class RolloutEngine:
         return base + [self.env.dt, p[r_key], p["speed_limit"],
                        p["dist2goal"], act_lim]
 
-    def _body(self):
+    def _body(self, explore: bool = False):
         data = GraphBatch(x=self.x, pos=self.states[:, :self.pos_dim],
                           states=self.states, edge_index=self.ei,
                           edge_attr=self.ea, agent_mask=self.agent_mask,
@@ -110,7 +110,14 @@ class RolloutEngine:
         data.seg_dst = self.seg
         data.agents_first_n = self.n if self.agent_mask is not None else None
         with torch.no_grad():
-            action = self.algo.actor(data) * self.explore
+            if explore:
+                # exploration step: the action is zeroed (pure nominal
+                # control, reference gcbf/algo/gcbf.py:131-132) — the
+                # reference still runs the actor and discards the output;
+                # this graph skips it entirely (observably identical)
+                action = self.zero_action
+            else:
+                action = self.algo.actor(data)
             unsafe_any = self.env.unsafe_mask(data).any()
             out = ops.env_step_fused(self.kind, *self._step_args(action))
             new_states, u_ref_next, reward, reach, collision = out
@@ -136,11 +143,17 @@ class RolloutEngine:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(3):
-                self._body()
+                self._body(False)
+            self._body(True)
         torch.cuda.current_stream().wait_stream(s)
-        self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
-            self._body()
+        # two graphs over the same buffers: policy steps run the actor,
+        # exploration steps (zeroed action) skip it entirely
+        self.g_policy = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_policy):
+            self._body(False)
+        self.g_explore = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_explore, pool=self.g_policy.pool()):
+            self._body(True)
         # restore pre-warmup state
         self.states.copy_(saved[0])
         self.u_ref.copy_(saved[1])
@@ -185,8 +198,10 @@ class RolloutEngine:
             agent_mask=self.agent_mask,
             u_ref=self.u_ref.clone())
 
-        self.explore.fill_(0.0 if np.random.rand() < prob else 1.0)
-        self.graph.replay()
+        if np.random.rand() < prob:
+            self.g_explore.replay()
+        else:
+            self.g_policy.replay()
 
         flags = self.flags.cpu()  # ONE host sync per step
         self.E = int(flags[0])
