@@ -558,3 +558,33 @@ def test_stress_config_n256_updates():
         if algo.is_update(step):
             out = algo.update(step, None)
     assert all(np.isfinite(v) for v in out.values())
+
+
+def test_segment_attn_deterministic():
+    """No atomics, fixed CSR reduction order: repeated runs are bitwise
+    identical (SURVEY.md §5.2 determinism requirement)."""
+    msg, gate, dst = _rand_edges(5000, 300, 256, seed=11)
+    out1 = ops.segment_attn_aggregate(msg, gate, dst, 300)
+    out2 = ops.segment_attn_aggregate(msg, gate, dst, 300)
+    assert torch.equal(out1, out2)
+    m = msg.clone().requires_grad_(True)
+    g = gate.clone().requires_grad_(True)
+    o = ops.segment_attn_aggregate(m, g, dst, 300)
+    go = torch.randn_like(o)
+    o.backward(go)
+    g1m, g1g = m.grad.clone(), g.grad.clone()
+    m.grad = None; g.grad = None
+    o = ops.segment_attn_aggregate(m, g, dst, 300)
+    o.backward(go)
+    assert torch.equal(m.grad, g1m)
+    assert torch.equal(g.grad, g1g)
+
+
+def test_fused_linear_deterministic():
+    from gcbf_amd import _C
+    A = torch.randn(512, 2048, device="cuda").bfloat16()
+    W = torch.randn(2048, 2048, device="cuda").bfloat16()
+    b = torch.randn(2048, device="cuda")
+    o1 = _C.fused_linear(A, W, b, 1, True)
+    o2 = _C.fused_linear(A, W, b, 1, True)
+    assert torch.equal(o1, o2)
