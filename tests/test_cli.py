@@ -83,3 +83,12 @@ def test_nominal_test_cli(tmp_path):
     r = _run([sys.executable, "test.py", "--env", "SimpleCar", "-n", "3",
               "--epi", "1", "--no-video", "--cpu"], timeout=1800)
     assert r.returncode == 0, r.stderr[-3000:]
+
+
+def test_plot_cbf_cli(trained_run):
+    r = _run([sys.executable, "plot_cbf.py", "--path", trained_run,
+              "--area-size", "4.0", "--epi", "1", "--max-steps", "2",
+              "--cpu"], timeout=1800)
+    assert r.returncode == 0, r.stderr[-3000:]
+    agent_dir = os.path.join(trained_run, "figs", "agent_0", "epi_0")
+    assert os.path.exists(os.path.join(agent_dir, "0.pdf"))
