@@ -75,6 +75,10 @@ def test(args):
     algo = make_algo(settings["algo"], env, num_agents, env.node_dim,
                      env.edge_dim, env.action_dim, device,
                      hyperparams=settings.get("hyper_params"))
+    dtype = args.dtype or ("bf16" if use_cuda else "fp32")
+    if dtype == "bf16" and use_cuda and settings["algo"] in ("gcbf", "macbf"):
+        from gcbf_amd.utils.amp import enable_bf16
+        enable_bf16(algo)
 
     if args.path is None:
         assert args.env is not None and args.num_agents is not None
@@ -178,4 +182,6 @@ if __name__ == "__main__":
     parser.add_argument("--rand", type=float, default=30)
     parser.add_argument("--seed", type=int, default=0)
     parser.add_argument("--cpu", action="store_true", default=False)
+    parser.add_argument("--dtype", type=str, default=None,
+                        choices=[None, "bf16", "fp32"])
     test(parser.parse_args())
