@@ -55,6 +55,8 @@ def run(args):
 
     params_env = make_env(args.env, args.num_agents, device).default_params
     params_env["num_obs"] = args.obs
+    if args.area_size is not None:
+        params_env["area_size"] = args.area_size
     env = make_env(args.env, args.num_agents, device, params=params_env)
     env.train()
     hyper = read_params(args.env, "gcbf")
@@ -175,6 +177,7 @@ if __name__ == "__main__":
     p.add_argument("--env", type=str, default="DubinsCar")
     p.add_argument("-n", "--num-agents", type=int, default=16)
     p.add_argument("--obs", type=int, default=0)
+    p.add_argument("--area-size", type=float, default=None)
     p.add_argument("--batch-size", type=int, default=512)
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--dtype", type=str, default=None,
