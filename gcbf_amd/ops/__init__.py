@@ -125,6 +125,30 @@ def segment_softmax(gate: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
     return eager.segment_softmax(gate, dst, num_nodes)
 
 
+class _SegmentMaxHip(torch.autograd.Function):
+    """CSR segment max on the HIP kernel (dst-sorted edges)."""
+
+    @staticmethod
+    def forward(ctx, values: Tensor, dst: Tensor, num_nodes: int):
+        ext = _require_ext("segment_max")
+        vdtype = values.dtype
+        v32 = values if vdtype == torch.float32 else values.float()
+        ptr = _csr_ptr(dst, num_nodes)
+        out, argmax = ext.segment_max_fwd(v32.contiguous(), ptr)
+        ctx.save_for_backward(argmax)
+        ctx.E = values.shape[0]
+        ctx.vdtype = vdtype
+        return out.to(vdtype)
+
+    @staticmethod
+    def backward(ctx, grad_out: Tensor):
+        (argmax,) = ctx.saved_tensors
+        from gcbf_amd import _C
+        dval = _C.segment_max_bwd(grad_out.float().contiguous(), argmax,
+                                  ctx.E)
+        return dval.to(ctx.vdtype), None, None
+
+
 class _SegmentMax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, values: Tensor, dst: Tensor, num_nodes: int):
@@ -157,6 +181,8 @@ class _SegmentMax(torch.autograd.Function):
 
 
 def segment_max(values: Tensor, dst: Tensor, num_nodes: int) -> Tensor:
+    if values.is_cuda and _load_ext() is not None:
+        return _SegmentMaxHip.apply(values, dst, num_nodes)
     return _SegmentMax.apply(values, dst, num_nodes)
 
 

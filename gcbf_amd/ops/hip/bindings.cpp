@@ -13,6 +13,12 @@ extern "C" void launch_seg_attn_bwd(const float* grad_out, const float* msg,
                                     const float* att, const int* ptr,
                                     float* dmsg, float* dgate, int N, int D,
                                     hipStream_t stream);
+extern "C" void launch_seg_max_fwd(const float* values, const int* ptr,
+                                   float* out, int* argmax, int N, int D,
+                                   hipStream_t stream);
+extern "C" void launch_seg_max_bwd(const float* grad_out, const int* argmax,
+                                   float* dval, int N, int D,
+                                   hipStream_t stream);
 extern "C" void launch_radius_count(const float* pos, int* counts, int B,
                                     int N, int n_rec, int P, float r,
                                     int topk, hipStream_t stream);
@@ -104,6 +110,35 @@ std::vector<torch::Tensor> segment_attn_bwd(torch::Tensor grad_out,
                             dmsg.data_ptr<float>(), dgate.data_ptr<float>(),
                             (int)N, (int)D, current_stream());
     return {dmsg, dgate};
+}
+
+std::vector<torch::Tensor> segment_max_fwd(torch::Tensor values,
+                                           torch::Tensor ptr) {
+    CHECK_IN(values);
+    CHECK_IN(ptr);
+    const int64_t D = values.size(1);
+    const int64_t N = ptr.size(0) - 1;
+    auto out = torch::empty({N, D}, values.options());
+    auto argmax = torch::empty({N, D},
+                               values.options().dtype(torch::kInt32));
+    if (N > 0)
+        launch_seg_max_fwd(values.data_ptr<float>(), ptr.data_ptr<int>(),
+                           out.data_ptr<float>(), argmax.data_ptr<int>(),
+                           (int)N, (int)D, current_stream());
+    return {out, argmax};
+}
+
+torch::Tensor segment_max_bwd(torch::Tensor grad_out, torch::Tensor argmax,
+                              int64_t E) {
+    CHECK_IN(grad_out);
+    CHECK_IN(argmax);
+    const int64_t N = grad_out.size(0), D = grad_out.size(1);
+    auto dval = torch::zeros({E, D}, grad_out.options());
+    if (N > 0)
+        launch_seg_max_bwd(grad_out.data_ptr<float>(),
+                           argmax.data_ptr<int>(), dval.data_ptr<float>(),
+                           (int)N, (int)D, current_stream());
+    return dval;
 }
 
 std::vector<torch::Tensor> build_graph(torch::Tensor pos,
@@ -303,6 +338,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused scatter-softmax + weighted scatter-sum (forward)");
     m.def("segment_attn_bwd", &segment_attn_bwd,
           "fused scatter-softmax + weighted scatter-sum (backward)");
+    m.def("segment_max_fwd", &segment_max_fwd,
+          "CSR segment max with argmax (forward)");
+    m.def("segment_max_bwd", &segment_max_bwd,
+          "CSR segment max scatter-back (backward)");
     m.def("build_graph", &build_graph,
           "batched dense radius graph + edge_attr (count/scan/fill)");
     m.def("build_graph_padded", &build_graph_padded,

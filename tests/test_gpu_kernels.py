@@ -588,3 +588,18 @@ def test_fused_linear_deterministic():
     o1 = _C.fused_linear(A, W, b, 1, True)
     o2 = _C.fused_linear(A, W, b, 1, True)
     assert torch.equal(o1, o2)
+
+
+def test_segment_max_hip_matches_eager():
+    msg, _, dst = _rand_edges(500, 40, 128, seed=5)
+    out = ops.segment_max(msg, dst, 40)
+    ref = eager.segment_max(msg.cpu(), dst.cpu(), 40)
+    assert torch.allclose(out.cpu(), ref, atol=1e-6)
+    # backward: gradient lands on the argmax edges only
+    m = msg.clone().requires_grad_(True)
+    o = ops.segment_max(m, dst, 40)
+    o.sum().backward()
+    m2 = msg.cpu().requires_grad_(True)
+    o2 = eager.segment_max(m2, dst.cpu(), 40)
+    o2.sum().backward()
+    assert torch.allclose(m.grad.cpu(), m2.grad, atol=1e-6)
