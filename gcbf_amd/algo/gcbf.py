@@ -17,7 +17,7 @@ MI355X redesign points:
 from __future__ import annotations
 
 import os
-from typing import Callable, List, Optional
+from typing import Callable, Optional
 
 import numpy as np
 import torch

@@ -23,7 +23,7 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
-import numpy as np
+
 import torch
 from torch import Tensor
 

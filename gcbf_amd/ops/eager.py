@@ -12,7 +12,7 @@ and the CPU execution path.  Semantics mirror the reference implementation:
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 from torch import Tensor

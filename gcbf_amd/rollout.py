@@ -23,7 +23,7 @@ without re-capture.
 """
 from __future__ import annotations
 
-from typing import Optional
+
 
 import numpy as np
 import torch
