@@ -41,7 +41,9 @@ def init_distributed(backend: Optional[str] = None) -> tuple:
     if world == 1:
         return rank, world, local
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get(
+            "GCBF_AMD_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo")
     if not dist.is_initialized():
         # container hostnames may not resolve — default to loopback
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
