@@ -47,7 +47,8 @@ def run(args):
     set_seed(args.seed + rank)
 
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", local_rank) if use_cuda \
+    device = torch.device(
+        "cuda", local_rank % torch.cuda.device_count()) if use_cuda \
         else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)

@@ -27,8 +27,9 @@ def train(args):
 
     use_cuda = torch.cuda.is_available() and not args.cpu
     if use_cuda:
-        device = torch.device("cuda", local_rank if world_size > 1
-                              else args.gpu)
+        dev_idx = (local_rank % torch.cuda.device_count()
+                   if world_size > 1 else args.gpu)
+        device = torch.device("cuda", dev_idx)
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
