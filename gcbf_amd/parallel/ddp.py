@@ -50,7 +50,9 @@ def init_distributed(backend: Optional[str] = None) -> tuple:
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
     if torch.cuda.is_available():
-        torch.cuda.set_device(local)
+        # modulo: lets multi-rank smoke tests share one GPU; a no-op on a
+        # real N-GPU node where local < device_count
+        torch.cuda.set_device(local % torch.cuda.device_count())
     return rank, world, local
 
 
