@@ -174,27 +174,31 @@ class GCBF(Algorithm):
             if prof:
                 t1 = _tick(); prof["fwd"] += t1 - t0; t0 = t1
 
-            # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177)
+            # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177).
+            # Masked means are computed as weighted sums: boolean-mask
+            # indexing (h[mask]) calls nonzero and forces a device→host
+            # sync per mask per inner iteration; the weighted form is
+            # mathematically identical (incl. the empty-mask fallbacks of
+            # loss 0 / acc 1) and stays on device.
             unsafe_mask = self._env.unsafe_mask(graphs)
-            h_unsafe = h[unsafe_mask]
-            if h_unsafe.numel():
-                loss_unsafe = torch.mean(torch.relu(h_unsafe + eps))
-                acc_unsafe = torch.mean(
-                    torch.less(h_unsafe, 0).type_as(h_unsafe))
-            else:
-                loss_unsafe = torch.tensor(0.0).type_as(h)
-                acc_unsafe = torch.tensor(1.0).type_as(h)
+            hv = h[:, 0]
+            wu = unsafe_mask.to(hv.dtype)
+            cu = wu.sum()
+            cu1 = cu.clamp(min=1)
+            any_u = (cu > 0).to(hv.dtype)
+            loss_unsafe = any_u * (torch.relu(hv + eps) * wu).sum() / cu1
+            acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
+                          + (1 - any_u))
 
             # safe region: h > 0
             safe_mask = self._env.safe_mask(graphs)
-            h_safe = h[safe_mask]
-            if h_safe.numel():
-                loss_safe = torch.mean(torch.relu(-h_safe + eps))
-                acc_safe = torch.mean(
-                    torch.greater_equal(h_safe, 0).type_as(h_safe))
-            else:
-                loss_safe = torch.tensor(0.0).type_as(h)
-                acc_safe = torch.tensor(1.0).type_as(h)
+            ws = safe_mask.to(hv.dtype)
+            cs = ws.sum()
+            cs1 = cs.clamp(min=1)
+            any_s = (cs > 0).to(hv.dtype)
+            loss_safe = any_s * (torch.relu(-hv + eps) * ws).sum() / cs1
+            acc_safe = (any_s * ((hv >= 0).to(hv.dtype) * ws).sum() / cs1
+                        + (1 - any_s))
 
             if prof:
                 t1 = _tick(); prof["mask"] += t1 - t0; t0 = t1

@@ -87,26 +87,26 @@ class MACBF(GCBF):
             h = self.cbf(graphs)
             actions = self.actor(graphs)
 
-            # per-edge masks (reference macbf.py:144, 156)
+            # per-edge masks (reference macbf.py:144, 156) as weighted
+            # sums — identical values without the nonzero host sync
+            hv = h[:, 0]
             unsafe_mask = self._env.unsafe_mask(graphs, return_edge=True)
-            h_unsafe = h[unsafe_mask]
-            if h_unsafe.numel():
-                loss_unsafe = torch.mean(torch.relu(h_unsafe + eps))
-                acc_unsafe = torch.mean(
-                    torch.less(h_unsafe, 0).type_as(h_unsafe))
-            else:
-                loss_unsafe = torch.tensor(0.0).type_as(h)
-                acc_unsafe = torch.tensor(1.0).type_as(h)
+            wu = unsafe_mask.to(hv.dtype)
+            cu = wu.sum()
+            cu1 = cu.clamp(min=1)
+            any_u = (cu > 0).to(hv.dtype)
+            loss_unsafe = any_u * (torch.relu(hv + eps) * wu).sum() / cu1
+            acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
+                          + (1 - any_u))
 
             safe_mask = self._env.safe_mask(graphs, return_edge=True)
-            h_safe = h[safe_mask]
-            if h_safe.numel():
-                loss_safe = torch.mean(torch.relu(-h_safe + eps))
-                acc_safe = torch.mean(
-                    torch.greater_equal(h_safe, 0).type_as(h_safe))
-            else:
-                loss_safe = torch.tensor(0.0).type_as(h)
-                acc_safe = torch.tensor(1.0).type_as(h)
+            ws = safe_mask.to(hv.dtype)
+            cs = ws.sum()
+            cs1 = cs.clamp(min=1)
+            any_s = (cs > 0).to(hv.dtype)
+            loss_safe = any_s * (torch.relu(-hv + eps) * ws).sum() / cs1
+            acc_safe = (any_s * ((hv >= 0).to(hv.dtype) * ws).sum() / cs1
+                        + (1 - any_s))
 
             # ḣ on the fixed topology only (reference macbf.py:168-173)
             graphs_next = self._env.forward_graph(graphs, actions)
