@@ -279,17 +279,14 @@ class SimpleCar(MultiAgentEnv):
         action = -torch.einsum("us,bns->bnu", self._K, diff)
         action = action.reshape(-1, self.action_dim)
 
-        # speed-limit penalty
+        # speed-limit penalty (branch-free: relu gates the over-speed rows,
+        # so no data-dependent host sync — reference simple_car.py:296-302)
         states = states.reshape(-1, self.state_dim)
-        speed = states[:, 2:].norm(dim=1)
-        over = speed - self._params["speed_limit"] > 0
-        if over.any():
-            v = states[over, 2:]
-            v_dir = v / v.norm(dim=1, keepdim=True)
-            action[over] = action[over] - (
-                v.norm(dim=1, keepdim=True) - self._params["speed_limit"]
-            ) * v_dir * 50
-        return action
+        v = states[:, 2:]
+        speed = v.norm(dim=1, keepdim=True)
+        penalty = torch.relu(speed - self._params["speed_limit"]) * 50
+        v_dir = v / speed.clamp(min=1e-12)
+        return action - penalty * v_dir
 
     # ----------------------------------------------------------------- masks
     def _pairwise(self, data: GraphBatch, diag_offset: float

@@ -275,16 +275,13 @@ class SimpleDrone(MultiAgentEnv):
         action = -torch.einsum("us,bns->bnu", self._K, diff)
         action = action.reshape(-1, self.action_dim)
 
+        # branch-free over-speed penalty (reference simple_drone.py:367-375)
         states = states.reshape(-1, self.state_dim)
-        speed = states[:, 3:].norm(dim=1)
-        over = speed - self._params["speed_limit"] > 0
-        if over.any():
-            v = states[over, 3:]
-            v_dir = v / v.norm(dim=1, keepdim=True)
-            action[over] = action[over] - (
-                v.norm(dim=1, keepdim=True) - self._params["speed_limit"]
-            ) * v_dir * 10
-        return action
+        v = states[:, 3:]
+        speed = v.norm(dim=1, keepdim=True)
+        penalty = torch.relu(speed - self._params["speed_limit"]) * 10
+        v_dir = v / speed.clamp(min=1e-12)
+        return action - penalty * v_dir
 
     # ----------------------------------------------------------------- masks
     def _pairwise_agent_rows(self, data: GraphBatch, diag_offset: float
