@@ -14,14 +14,17 @@ import torch.multiprocessing as mp
 from gcbf_amd.parallel import GradSynchronizer, broadcast_modules
 
 
+BASE_PORT = 29000 + (os.getpid() % 900)
+
+
 def _init(rank, world, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
 
 
-def _worker_grad_sync(rank, world, q):
-    _init(rank, world, 29511)
+def _worker_grad_sync(rank, world, q, port):
+    _init(rank, world, port)
     torch.manual_seed(rank)  # different grads per rank
     m = torch.nn.Linear(4, 3)
     # identical weights
@@ -40,7 +43,7 @@ def _worker_grad_sync(rank, world, q):
 def test_grad_synchronizer_averages():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_worker_grad_sync, args=(r, 2, q))
+    ps = [ctx.Process(target=_worker_grad_sync, args=(r, 2, q, BASE_PORT))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -56,8 +59,8 @@ def test_grad_synchronizer_averages():
     assert torch.allclose(results[1][1], mean, atol=1e-6)
 
 
-def _worker_gcbf_dp(rank, world, q):
-    _init(rank, world, 29519)
+def _worker_gcbf_dp(rank, world, q, port):
+    _init(rank, world, port)
     from gcbf_amd.algo import make_algo
     from gcbf_amd.env import make_env
     from gcbf_amd.trainer.utils import set_seed
@@ -88,7 +91,8 @@ def _worker_gcbf_dp(rank, world, q):
 def test_gcbf_dp_two_ranks_stay_in_sync():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_worker_gcbf_dp, args=(r, 2, q))
+    ps = [ctx.Process(target=_worker_gcbf_dp,
+                      args=(r, 2, q, BASE_PORT + 7))
           for r in range(2)]
     for p in ps:
         p.start()
