@@ -148,12 +148,17 @@ extern "C" __global__ void radius_fill(
         const unsigned long long mask = __ballot(pred);
         if (pred) {
             const int off = base + __popcll(mask & ((1ull << lane) - 1ull));
-            const long src = gbase + j;
-            const long dst = gbase + i;
-            edge_index[off] = src;
-            edge_index[E_total + off] = dst;
-            write_attr(edge_attr + (size_t)off * A, states, S, attr_kind,
-                       src, dst);
+            // guard: with a soft capacity (padded engine buffers) the edge
+            // count can exceed E_total — skip writes; the host detects the
+            // overflow from the published count and redoes the step eagerly
+            if (off < E_total) {
+                const long src = gbase + j;
+                const long dst = gbase + i;
+                edge_index[off] = src;
+                edge_index[E_total + off] = dst;
+                write_attr(edge_attr + (size_t)off * A, states, S, attr_kind,
+                           src, dst);
+            }
         }
         base += __popcll(mask);
     }

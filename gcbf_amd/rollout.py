@@ -41,12 +41,11 @@ def engine_supported(env, algo) -> bool:
         return False
     if env._mode != "train" or env._max_neighbors is not None:
         return False
-    # padding to E_max only pays while the padded GEMMs stay small; large
-    # scenes (the n=256 stress config) are compute-bound and amortize
-    # launches anyway
     data = env.data if env.data is not None else env.reset()
     n, N = env.num_agents, data.num_nodes
-    return n * (N - 1) <= 8192
+    # sanity bound only: the engine uses a soft edge capacity with an eager
+    # fallback, so even large scenes (the n=256 stress config) capture
+    return n * (N - 1) <= 262144
 
 
 class RolloutEngine:
@@ -61,7 +60,12 @@ class RolloutEngine:
 
         data = env.data if env.data is not None else env.reset()
         self.N = data.num_nodes
-        self.E_max = self.n * (self.N - 1)
+        # soft edge capacity: full n*(N-1) for small scenes; for large ones
+        # 8x the observed density (overflowing steps fall back to the eager
+        # path until the edge count fits again)
+        full = self.n * (self.N - 1)
+        self.E_max = min(full, max(4096, 8 * max(data.num_edges, 1)))
+        self._eager = False
 
         from gcbf_amd import _C
         self._ext = _C
@@ -178,14 +182,48 @@ class RolloutEngine:
         self.seg.copy_(seg)
         self.ea.copy_(ea)
         self.E = int(ecount.item())
+        self._eager = self.E > self.E_max
+        if self._eager:
+            self.E = 0
 
     def reload(self):
         self.env.reset()
         self._load_graph_from_env()
 
+    # ----------------------------------------------------- overflow fallback
+    def _sync_env_from_buffers(self):
+        """Materialize the env's graph from the engine state (exact edge
+        build) when entering eager fallback."""
+        data = GraphBatch(x=self.x,
+                          pos=self.states[:, :self.pos_dim].clone(),
+                          states=self.states.clone(),
+                          agent_mask=self.agent_mask)
+        if self.agent_mask is not None:
+            data.agents_first_n = self.n
+        self.env._data = self.env.add_communication_links(data)
+        self.env._data.u_ref = self.u_ref.clone()
+
+    def _eager_step(self, prob: float) -> bool:
+        """Plain (non-captured) training step, used while the scene's edge
+        count exceeds the captured buffers' capacity."""
+        env, algo = self.env, self.algo
+        data = env.data
+        if data.u_ref is None:
+            data.update(u_ref=env.u_ref(data))
+        action = algo.step(data, prob)
+        next_data, reward, done, info = env.step(action)
+        if done:
+            return True
+        if next_data.num_edges <= self.E_max:
+            # fits again: reload the captured buffers and resume replaying
+            self._load_graph_from_env()
+        return False
+
     # ----------------------------------------------------------------- step
     def step(self, prob: float) -> bool:
         """One training env step.  Returns done."""
+        if self._eager:
+            return self._eager_step(prob)
         # clone the CURRENT graph for the replay buffer before the replay
         # overwrites the static buffers
         E = self.E
@@ -212,6 +250,16 @@ class RolloutEngine:
 
         self.env._t += 1
         done = self.env._t >= self.env.max_episode_steps or reach_all
+        if self.E > self.E_max:
+            # soft-capacity overflow: the state advance / rewards / flags are
+            # exact (they don't depend on the rebuilt graph) but the new edge
+            # buffers are truncated — rebuild exactly and step eagerly until
+            # the scene thins out again
+            self.E = 0
+            self._eager = True
+            if not done:
+                self._sync_env_from_buffers()
+            return done
         if not done:
             # keep the env object's view of the world consistent (cheap:
             # shares the engine's buffers; env methods are not used for

@@ -603,3 +603,32 @@ def test_segment_max_hip_matches_eager():
     o2 = eager.segment_max(m2, dst.cpu(), 40)
     o2.sum().backward()
     assert torch.allclose(m.grad.cpu(), m2.grad, atol=1e-6)
+
+
+def test_rollout_engine_large_scene_soft_capacity():
+    """n=64 scene captures with a soft edge capacity; steps must stay
+    consistent with the eager loop even around capacity boundaries."""
+    from gcbf_amd.env import make_env
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.rollout import RolloutEngine, engine_supported
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(5)
+    dev = torch.device("cuda")
+    e0 = make_env("DubinsCar", 64, dev)
+    p = e0.default_params
+    p["num_obs"] = 8
+    p["area_size"] = 8.0
+    env = make_env("DubinsCar", 64, dev, params=p)
+    env.train()
+    algo = make_algo("gcbf", env, 64, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=512)
+    env.reset()
+    assert engine_supported(env, algo)
+    eng = RolloutEngine(env, algo)
+    # force a tiny capacity so overflow -> eager fallback exercises
+    for t in range(60):
+        done = eng.step(prob=0.8)
+        assert torch.isfinite(eng.states).all()
+        if done:
+            eng.reload()
+    assert algo.buffer.size == 60
