@@ -50,7 +50,7 @@ def engine_supported(env, algo) -> bool:
 
 class RolloutEngine:
 
-    def __init__(self, env, algo):
+    def __init__(self, env, algo, edge_capacity: int | None = None):
         self.env = env
         self.algo = algo
         self.device = env.device
@@ -64,7 +64,10 @@ class RolloutEngine:
         # 8x the observed density (overflowing steps fall back to the eager
         # path until the edge count fits again)
         full = self.n * (self.N - 1)
-        self.E_max = min(full, max(4096, 8 * max(data.num_edges, 1)))
+        if edge_capacity is not None:  # test hook
+            self.E_max = min(full, edge_capacity)
+        else:
+            self.E_max = min(full, max(4096, 8 * max(data.num_edges, 1)))
         self._eager = False
 
         from gcbf_amd import _C

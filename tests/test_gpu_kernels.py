@@ -605,9 +605,10 @@ def test_segment_max_hip_matches_eager():
     assert torch.allclose(m.grad.cpu(), m2.grad, atol=1e-6)
 
 
-def test_rollout_engine_large_scene_soft_capacity():
-    """n=64 scene captures with a soft edge capacity; steps must stay
-    consistent with the eager loop even around capacity boundaries."""
+def test_rollout_engine_soft_capacity_overflow_fallback():
+    """Soft-capacity engine: with an edge capacity barely above the initial
+    edge count, steps must overflow into the eager fallback and later resume
+    captured replay, with the buffer/state staying consistent throughout."""
     from gcbf_amd.env import make_env
     from gcbf_amd.algo import make_algo
     from gcbf_amd.rollout import RolloutEngine, engine_supported
@@ -622,13 +623,21 @@ def test_rollout_engine_large_scene_soft_capacity():
     env.train()
     algo = make_algo("gcbf", env, 64, env.node_dim, env.edge_dim,
                      env.action_dim, dev, batch_size=512)
-    env.reset()
+    data = env.reset()
     assert engine_supported(env, algo)
-    eng = RolloutEngine(env, algo)
-    # force a tiny capacity so overflow -> eager fallback exercises
-    for t in range(60):
-        done = eng.step(prob=0.8)
+    eng = RolloutEngine(env, algo, edge_capacity=data.num_edges + 8)
+    saw_eager = saw_captured = False
+    for t in range(120):
+        was_eager = eng._eager
+        saw_eager |= was_eager
+        saw_captured |= not was_eager
+        done = eng.step(prob=0.5)
         assert torch.isfinite(eng.states).all()
+        if not eng._eager:
+            assert torch.isfinite(eng.ea[: eng.E]).all()
         if done:
             eng.reload()
-    assert algo.buffer.size == 60
+    assert algo.buffer.size == 120
+    assert saw_captured
+    # with capacity this tight the moving scene must overflow at least once
+    assert saw_eager
