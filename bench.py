@@ -73,7 +73,8 @@ def run(args):
 
     total_schedule = 500_000  # exploration schedule of the paper config
     data = env.reset()
-    prof = {"rollout_s": 0.0, "update_s": 0.0, "updates": 0}
+    prof = {"rollout_s": 0.0, "update_s": 0.0, "updates": 0,
+            "reset_s": 0.0, "resets": 0}
 
     engine = None
     if use_cuda and not args.no_capture:
@@ -96,7 +97,11 @@ def run(args):
         if engine is not None:
             done = engine.step(prob=1 - (step - 1) / total_schedule)
             if done:
+                tr = time.perf_counter() if timed else 0.0
                 engine.reload()
+                if timed:
+                    prof["reset_s"] += time.perf_counter() - tr
+                    prof["resets"] += 1
         else:
             data.update(u_ref=env.u_ref(data))
             action = algo.step(data, prob=1 - (step - 1) / total_schedule)
@@ -129,10 +134,12 @@ def run(args):
     elapsed = time.perf_counter() - t0
     if args.profile and rank == 0:
         import sys
-        print(f"# profile: rollout {prof['rollout_s']:.2f}s "
-              f"({prof['rollout_s'] / max(args.steps - prof['updates'], 1) * 1e3:.2f} ms/step), "
+        roll = prof["rollout_s"] - prof["reset_s"]
+        print(f"# profile: rollout {roll:.2f}s "
+              f"({roll / max(args.steps - prof['updates'], 1) * 1e3:.2f} ms/step), "
               f"update {prof['update_s']:.2f}s over {prof['updates']} updates "
-              f"({prof['update_s'] / max(prof['updates'], 1):.2f} s/update)",
+              f"({prof['update_s'] / max(prof['updates'], 1):.2f} s/update), "
+              f"resets {prof['reset_s']:.2f}s over {prof['resets']}",
               file=sys.stderr, flush=True)
 
     # max over ranks
