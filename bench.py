@@ -124,6 +124,13 @@ def run(args):
     # ---- warmup (untimed) ----
     for step in range(1, args.warmup + 1):
         one_step(step)
+    if (args.warmup < args.batch_size
+            and algo.buffer.size >= args.batch_size // 5):
+        # no update fell inside the warmup phase, so the FIRST timed update
+        # would pay the one-time hipBLASLt algorithm-search + autograd-warmup
+        # costs (~0.4 s, measured).  Run one untimed update to absorb them;
+        # the timed region still performs every update it owes.
+        algo.update(args.batch_size, None)
 
     # ---- timed region ----
     _barrier_sync(device)
