@@ -28,6 +28,9 @@ class Buffer:
         # at sample/merge time (the reference syncs every rollout step,
         # gcbf/algo/gcbf.py:133-137)
         self._pending: List[tuple] = []
+        # optional hook: called with each appended graph (the captured
+        # update engine uses it to mirror states into its device ring)
+        self.on_append = None
 
     @property
     def data(self) -> List[GraphBatch]:
@@ -47,6 +50,8 @@ class Buffer:
         self._pending.clear()
 
     def append(self, data: GraphBatch, is_safe):
+        if self.on_append is not None:
+            self.on_append(data)
         self._data.append(data)
         import torch
         if torch.is_tensor(is_safe):

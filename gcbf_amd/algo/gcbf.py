@@ -105,6 +105,11 @@ class GCBF(Algorithm):
         # DP hook: called after backward, before the optimizer steps
         self.grad_sync: Optional[Callable[[], None]] = None
 
+        # hipGraph-captured update engine (created lazily at first update
+        # when supported); None -> eager inner iterations
+        self._upd_engine = None
+        self._upd_engine_tried = False
+
     # ---------------------------------------------------------------- acting
     @torch.no_grad()
     def act(self, data: GraphBatch) -> Tensor:
@@ -125,24 +130,49 @@ class GCBF(Algorithm):
         return step % self.batch_size == 0
 
     # -------------------------------------------------------------- training
+    def _make_update_engine(self):
+        """Try to build the hipGraph-captured update engine (GPU only)."""
+        self._upd_engine_tried = True
+        if os.environ.get("GCBF_AMD_UPDATE_CAPTURE", "1") == "0":
+            return
+        if type(self) is not GCBF:
+            return
+        if self.device.type != "cuda":
+            return
+        env = self._env
+        data = env.data
+        if (data is None or data.agent_mask is not None
+                or getattr(env, "_max_neighbors", None) is not None):
+            return
+        try:
+            from ..update_engine import UpdateEngine
+            self._upd_engine = UpdateEngine(self, env)
+        except Exception as e:  # fall back to eager iterations
+            import warnings
+            warnings.warn(f"update capture unavailable ({e}); eager updates")
+            self._upd_engine = None
+
     def update(self, step: int, writer=None) -> dict:
         seg_len = 3
         inner_iter = self.params["inner_iter"]
-        eps = self.params["eps"]
-        alpha = self.params["alpha"]
-        acc_safe = acc_unsafe = acc_h_dot = torch.zeros((), dtype=torch.float)
         logs = []  # deferred scalars, synced once at the end
 
         prof = None
         if os.environ.get("GCBF_AMD_UPDATE_PROF") == "1":
             import time as _time
             prof = {"sample": 0.0, "batch": 0.0, "fwd": 0.0, "mask": 0.0,
-                    "hdot": 0.0, "bwd": 0.0, "opt": 0.0}
+                    "hdot": 0.0, "bwd": 0.0, "opt": 0.0, "engine": 0.0}
 
             def _tick():
                 if self.device.type == "cuda":
                     torch.cuda.synchronize()
                 return _time.perf_counter()
+        else:
+            _tick = None
+
+        if self._upd_engine is None and not self._upd_engine_tried:
+            self._make_update_engine()
+
         for i_inner in range(inner_iter):
             t0 = _tick() if prof else 0
             # sample segments from the current buffer and the replay memory
@@ -156,106 +186,126 @@ class GCBF(Algorithm):
             if prof:
                 t1 = _tick(); prof["sample"] += t1 - t0; t0 = t1
 
-            graphs = GraphBatch.from_list(graph_list)
-            graphs.edge_attr.requires_grad_(True)
-            if prof:
-                t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1
-            with trace_range("gcbf/forward"):
-                actions = self.actor(graphs)
-                # h and h_next in ONE doubled-batch CBF forward: halves the
-                # CBF forward/backward chains vs. the reference's separate
-                # calls (gcbf/algo/gcbf.py:161,194) and evaluates both sides
-                # of the finite difference under the SAME spectral-norm σ
-                graphs_next = self._env.forward_graph(graphs, actions)
-                both = GraphBatch.from_list([graphs, graphs_next])
-                h_both = self.cbf(both)
-                n_ag = h_both.shape[0] // 2
-                h, h_next = h_both[:n_ag], h_both[n_ag:]
-            if prof:
-                t1 = _tick(); prof["fwd"] += t1 - t0; t0 = t1
-
-            # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177).
-            # Masked means are computed as weighted sums: boolean-mask
-            # indexing (h[mask]) calls nonzero and forces a device→host
-            # sync per mask per inner iteration; the weighted form is
-            # mathematically identical (incl. the empty-mask fallbacks of
-            # loss 0 / acc 1) and stays on device.
-            unsafe_mask = self._env.unsafe_mask(graphs)
-            hv = h[:, 0]
-            wu = unsafe_mask.to(hv.dtype)
-            cu = wu.sum()
-            cu1 = cu.clamp(min=1)
-            any_u = (cu > 0).to(hv.dtype)
-            loss_unsafe = any_u * (torch.relu(hv + eps) * wu).sum() / cu1
-            acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
-                          + (1 - any_u))
-
-            # safe region: h > 0
-            safe_mask = self._env.safe_mask(graphs)
-            ws = safe_mask.to(hv.dtype)
-            cs = ws.sum()
-            cs1 = cs.clamp(min=1)
-            any_s = (cs > 0).to(hv.dtype)
-            loss_safe = any_s * (torch.relu(-hv + eps) * ws).sum() / cs1
-            acc_safe = (any_s * ((hv >= 0).to(hv.dtype) * ws).sum() / cs1
-                        + (1 - any_s))
-
-            if prof:
-                t1 = _tick(); prof["mask"] += t1 - t0; t0 = t1
-            # ḣ condition with the re-link residue trick
-            # (reference gcbf/algo/gcbf.py:191-209): the VALUE reflects the
-            # re-linked next graph, the GRADIENT flows through the
-            # fixed-topology path.
-            with trace_range("gcbf/h_dot"):
-                with torch.no_grad():
-                    relinked = self._env.add_communication_links_batched(
-                        graphs_next.detach())
-                    h_next_new_link = self.cbf(relinked)
-            h_dot = (h_next - h) / self._env.dt
-            h_dot_new_link = (h_next_new_link - h) / self._env.dt
-            residue = (h_dot_new_link - h_dot).detach()
-            h_dot = residue + h_dot
-
-            if prof:
-                t1 = _tick(); prof["hdot"] += t1 - t0; t0 = t1
-            loss_h_dot = torch.mean(torch.relu(-h_dot - alpha * h + eps))
-            acc_h_dot = torch.mean(
-                torch.greater_equal(h_dot + alpha * h, 0).type_as(h_dot))
-
-            loss_action = torch.mean(torch.square(actions).sum(dim=1))
-
-            loss = (self.params["loss_unsafe_coef"] * loss_unsafe +
-                    self.params["loss_safe_coef"] * loss_safe +
-                    self.params["loss_h_dot_coef"] * loss_h_dot +
-                    self.params["loss_action_coef"] * loss_action)
-
-            self.optim_cbf.zero_grad(set_to_none=True)
-            self.optim_actor.zero_grad(set_to_none=True)
-            with trace_range("gcbf/backward"):
-                loss.backward()
-            if self.grad_sync is not None:
-                with trace_range("gcbf/grad_allreduce"):
-                    self.grad_sync()
-            if prof:
-                t1 = _tick(); prof["bwd"] += t1 - t0; t0 = t1
-            with trace_range("gcbf/optim"):
-                torch.nn.utils.clip_grad_norm_(self.cbf.parameters(), 1e-3)
-                torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
-                self.optim_cbf.step()
-                self.optim_actor.step()
-
-            logs.append(torch.stack([
-                loss_unsafe.detach(), loss_safe.detach(),
-                loss_h_dot.detach(), loss_action.detach(),
-                acc_unsafe.detach(), acc_safe.detach(), acc_h_dot.detach()]))
+            log7 = None
+            if self._upd_engine is not None:
+                log7 = self._upd_engine.try_iter(graph_list)
+                if prof and log7 is not None:
+                    t1 = _tick(); prof["engine"] += t1 - t0; t0 = t1
+            if log7 is None:
+                log7 = self._iter_eager(graph_list, prof, _tick)
+            logs.append(log7)
 
         if prof:
-            t1 = _tick(); prof["opt"] += t1 - t0
             import sys
             print("# update prof: " + " ".join(
                 f"{k}={v * 1000:.1f}ms" for k, v in prof.items()),
                 file=sys.stderr, flush=True)
 
+        return self._update_tail(step, writer, logs, inner_iter)
+
+    def _iter_eager(self, graph_list, prof=None, _tick=None) -> Tensor:
+        """One eager inner iteration (sampled batch -> losses -> backward ->
+        optimizer).  Returns the 7-scalar log stack (device tensor)."""
+        eps = self.params["eps"]
+        alpha = self.params["alpha"]
+        t0 = _tick() if prof else 0
+
+        graphs = GraphBatch.from_list(graph_list)
+        graphs.edge_attr.requires_grad_(True)
+        if prof:
+            t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1
+        with trace_range("gcbf/forward"):
+            actions = self.actor(graphs)
+            # h and h_next in ONE doubled-batch CBF forward: halves the
+            # CBF forward/backward chains vs. the reference's separate
+            # calls (gcbf/algo/gcbf.py:161,194) and evaluates both sides
+            # of the finite difference under the SAME spectral-norm σ
+            graphs_next = self._env.forward_graph(graphs, actions)
+            both = GraphBatch.from_list([graphs, graphs_next])
+            h_both = self.cbf(both)
+            n_ag = h_both.shape[0] // 2
+            h, h_next = h_both[:n_ag], h_both[n_ag:]
+        if prof:
+            t1 = _tick(); prof["fwd"] += t1 - t0; t0 = t1
+
+        # unsafe region: h < 0 (reference gcbf/algo/gcbf.py:167-177).
+        # Masked means are computed as weighted sums: boolean-mask
+        # indexing (h[mask]) calls nonzero and forces a device→host
+        # sync per mask per inner iteration; the weighted form is
+        # mathematically identical (incl. the empty-mask fallbacks of
+        # loss 0 / acc 1) and stays on device.
+        unsafe_mask = self._env.unsafe_mask(graphs)
+        hv = h[:, 0]
+        wu = unsafe_mask.to(hv.dtype)
+        cu = wu.sum()
+        cu1 = cu.clamp(min=1)
+        any_u = (cu > 0).to(hv.dtype)
+        loss_unsafe = any_u * (torch.relu(hv + eps) * wu).sum() / cu1
+        acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
+                      + (1 - any_u))
+
+        # safe region: h > 0
+        safe_mask = self._env.safe_mask(graphs)
+        ws = safe_mask.to(hv.dtype)
+        cs = ws.sum()
+        cs1 = cs.clamp(min=1)
+        any_s = (cs > 0).to(hv.dtype)
+        loss_safe = any_s * (torch.relu(-hv + eps) * ws).sum() / cs1
+        acc_safe = (any_s * ((hv >= 0).to(hv.dtype) * ws).sum() / cs1
+                    + (1 - any_s))
+
+        if prof:
+            t1 = _tick(); prof["mask"] += t1 - t0; t0 = t1
+        # ḣ condition with the re-link residue trick
+        # (reference gcbf/algo/gcbf.py:191-209): the VALUE reflects the
+        # re-linked next graph, the GRADIENT flows through the
+        # fixed-topology path.
+        with trace_range("gcbf/h_dot"):
+            with torch.no_grad():
+                relinked = self._env.add_communication_links_batched(
+                    graphs_next.detach())
+                h_next_new_link = self.cbf(relinked)
+        h_dot = (h_next - h) / self._env.dt
+        h_dot_new_link = (h_next_new_link - h) / self._env.dt
+        residue = (h_dot_new_link - h_dot).detach()
+        h_dot = residue + h_dot
+
+        if prof:
+            t1 = _tick(); prof["hdot"] += t1 - t0; t0 = t1
+        loss_h_dot = torch.mean(torch.relu(-h_dot - alpha * h + eps))
+        acc_h_dot = torch.mean(
+            torch.greater_equal(h_dot + alpha * h, 0).type_as(h_dot))
+
+        loss_action = torch.mean(torch.square(actions).sum(dim=1))
+
+        loss = (self.params["loss_unsafe_coef"] * loss_unsafe +
+                self.params["loss_safe_coef"] * loss_safe +
+                self.params["loss_h_dot_coef"] * loss_h_dot +
+                self.params["loss_action_coef"] * loss_action)
+
+        self.optim_cbf.zero_grad(set_to_none=True)
+        self.optim_actor.zero_grad(set_to_none=True)
+        with trace_range("gcbf/backward"):
+            loss.backward()
+        if self.grad_sync is not None:
+            with trace_range("gcbf/grad_allreduce"):
+                self.grad_sync()
+        if prof:
+            t1 = _tick(); prof["bwd"] += t1 - t0; t0 = t1
+        with trace_range("gcbf/optim"):
+            torch.nn.utils.clip_grad_norm_(self.cbf.parameters(), 1e-3)
+            torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
+            self.optim_cbf.step()
+            self.optim_actor.step()
+
+        if prof:
+            t1 = _tick(); prof["opt"] += t1 - t0
+        return torch.stack([
+            loss_unsafe.detach(), loss_safe.detach(),
+            loss_h_dot.detach(), loss_action.detach(),
+            acc_unsafe.detach(), acc_safe.detach(), acc_h_dot.detach()])
+
+    def _update_tail(self, step, writer, logs, inner_iter) -> dict:
         # one host sync for the whole update's scalars
         log_vals = torch.stack(logs).cpu()
         if writer is not None:

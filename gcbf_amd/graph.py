@@ -43,7 +43,8 @@ class GraphBatch:
     """
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
-                 "u_ref", "_ptr", "_dst_ptr", "seg_dst", "agents_first_n")
+                 "u_ref", "_ptr", "_dst_ptr", "seg_dst", "agents_first_n",
+                 "ring_id")
 
     def __init__(
             self,
@@ -75,6 +76,9 @@ class GraphBatch:
         # first) — lets models use a static slice instead of boolean-mask
         # indexing, which is data-dependent and not hipGraph-capturable
         self.agents_first_n = None
+        # optional: slot id in the update engine's device ring (stamped by
+        # Buffer.on_append when the captured update engine is active)
+        self.ring_id = None
 
     # ------------------------------------------------------------------ sizes
     @property

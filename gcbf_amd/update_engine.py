@@ -1,0 +1,326 @@
+"""hipGraph-captured training update engine.
+
+At small scenes (the headline DubinsCar n=16 config) one eager inner
+update iteration costs ~15 ms of which most is kernel-launch/Python gaps:
+~300 tiny graphs are re-batched in Python, then ~200 small kernels run for
+the four GNN forwards, the losses and the backward.  This engine replays
+the whole iteration as two hipGraphs over fixed-capacity buffers:
+
+    graph F:  gather sampled graphs from a device ring (index_select)
+              → padded batched radius-graph build (fixed E_cap)
+              → actor forward → env.forward_graph (next states/attrs)
+    eager  :  exact re-link build + CBF forward of the re-linked next
+              graph (the ḣ residue trick needs EXACT edges — this stays
+              out of capture so overflow cannot corrupt gradients)
+    graph B:  doubled-batch CBF forward (h and h_next under one σ)
+              → weighted-sum losses → torch.autograd.grad
+    eager  :  grad all-reduce (DP) + clip + Adam + bf16 mirror refresh
+
+Design points:
+* The replay buffer's graphs are mirrored into a device ring of states +
+  u_ref at append time (``Buffer.on_append``).  Node features x and the
+  agent layout are static for a given env config, and edges are an exact
+  deterministic function of positions (the same HIP builder that built
+  them at rollout time), so the ring fully reconstructs any sampled graph.
+* Static shapes come from padding: the graph count is padded to G_cap
+  (3 · batch_size/5, the sampler's maximum) with weight-0 copies of a real
+  graph, and the edge buffers hold E_cap entries with the same
+  sentinel-segment scheme as the rollout engine.  Every loss/accuracy mean
+  becomes a weighted sum, which is exactly equal on the real rows.
+* Gradients come from ``torch.autograd.grad`` captured in graph B; the
+  resulting tensors are assigned to ``param.grad`` once, so the eager
+  optimizer tail (and the DP flat-buffer all-reduce) read refreshed
+  contents each replay.  Optimizer, clipping and communication stay
+  eager — no capturable-optimizer or captured-collective requirements,
+  and the engine composes with data parallelism unchanged.
+* Overflow of E_cap is detected from the published edge count AFTER
+  graph F (which touches no weights); the iteration falls back to the
+  exact eager path (``GCBF._iter_eager``).
+
+Reference behavior preserved (gcbf/algo/gcbf.py:140-230): sampling,
+balanced replay, the ḣ re-link residue trick, loss forms and coefficients,
+and the u_ref-from-current-goal behavior of replayed graphs
+(``forward_graph`` reads the env's live goal, kept address-stable here).
+"""
+from __future__ import annotations
+
+import torch
+
+from .graph import GraphBatch
+
+
+class UpdateEngine:
+    WARMUP_ITERS = 3
+
+    def __init__(self, algo, env):
+        from gcbf_amd import _C
+        self._ext = _C
+        self.algo = algo
+        self.env = env
+        self.device = env.device
+
+        data = env.data
+        assert data is not None and data.agent_mask is None
+        self.N = data.num_nodes            # nodes per graph (= agents here)
+        self.S = env.state_dim
+        self.pd = 3 if env.state_dim == 6 else 2
+        self.nd = env.node_dim
+        self.ad = env.action_dim
+        self.ed = env.edge_dim
+
+        seg_len = 3
+        self.G_cap = seg_len * (algo.batch_size // 5)
+        self.Ntot = self.G_cap * self.N
+        if algo.buffer.size == 0:
+            raise RuntimeError("empty buffer at engine init")
+
+        # edge capacity: 4x the replay window's mean edge count (padded
+        # GEMM rows are wasted work, so don't pad to the dense maximum)
+        mean_e = max(1.0, sum(g.num_edges for g in algo.buffer.data)
+                     / algo.buffer.size)
+        full = self.G_cap * self.N * (self.N - 1)
+        want = int(4.0 * mean_e * self.G_cap) + 2048
+        self.E_cap = min(full, (want + 2047) // 2048 * 2048)
+
+        dev = self.device
+        # ---- device ring of (states, u_ref) per appended graph
+        self.CAP = algo.buffer.MAX_SIZE + 2 * algo.batch_size
+        self.ring_states = torch.zeros(self.CAP, self.N, self.S, device=dev)
+        self.ring_uref = torch.zeros(self.CAP, self.N, self.ad, device=dev)
+        self.next_id = 0
+
+        # ---- static tiles / batch skeleton
+        self.x_tile = data.x.repeat(self.G_cap, 1).contiguous()
+        self.x2_tile = data.x.repeat(2 * self.G_cap, 1).contiguous()
+        self.ptr = torch.arange(self.G_cap + 1, dtype=torch.long,
+                                device=dev) * self.N
+        self.ptr2 = torch.arange(2 * self.G_cap + 1, dtype=torch.long,
+                                 device=dev) * self.N
+
+        # ---- per-iteration inputs (content updated before each replay)
+        self.idx_host = torch.empty(self.G_cap, dtype=torch.long,
+                                    pin_memory=True)
+        self.w_host = torch.empty(self.G_cap, pin_memory=True)
+        self.idx_dev = torch.zeros(self.G_cap, dtype=torch.long, device=dev)
+        self.w_dev = torch.zeros(self.G_cap, device=dev)
+
+        self.h_new = torch.zeros(self.Ntot, 1, device=dev)
+
+        # goal must be address-stable: forward_graph reads env._goal (the
+        # reference's replayed-graph u_ref uses the CURRENT goal)
+        self.goal_static = env._goal.clone().contiguous()
+        env._goal = self.goal_static
+
+        # backfill graphs appended before the engine existed, then hook
+        for g in list(algo.buffer.data) + list(algo.memory.data):
+            self.push(g)
+        algo.buffer.on_append = self.push
+
+        self._build()
+
+    # ------------------------------------------------------------- ring
+    def push(self, g: GraphBatch):
+        if g.ring_id is not None:
+            return
+        slot = self.next_id % self.CAP
+        self.ring_states[slot].copy_(g.states, non_blocking=True)
+        self.ring_uref[slot].copy_(g.u_ref, non_blocking=True)
+        g.ring_id = self.next_id
+        self.next_id += 1
+
+    # ------------------------------------------------------------- bodies
+    def _front(self):
+        """Gather + padded build + actor forward + env.forward_graph."""
+        env = self.env
+        with torch.no_grad():
+            nodes = self.ring_states.index_select(
+                0, self.idx_dev).reshape(self.Ntot, self.S)
+            uref = self.ring_uref.index_select(
+                0, self.idx_dev).reshape(self.Ntot, self.ad)
+            ei, seg, ea, ecount = self._ext.build_graph_padded(
+                nodes[:, :self.pd].contiguous(), nodes, self.G_cap, self.N,
+                env.params["comm_radius"], -1, env._attr_kind, self.ed,
+                self.E_cap)
+        gcur = GraphBatch(x=self.x_tile, pos=nodes[:, :self.pd],
+                          states=nodes, edge_index=ei, edge_attr=ea,
+                          u_ref=uref, ptr=self.ptr)
+        gcur.seg_dst = seg
+        actions = self.algo.actor(gcur)
+        gnext = env.forward_graph(gcur, actions)
+        gnext.seg_dst = seg
+        return gcur, gnext, actions, ecount
+
+    def _middle(self, gnext):
+        """EXACT re-link + CBF forward of the re-linked next graph (eager;
+        the residue needs exact edges, and overflow must not be able to
+        corrupt gradients)."""
+        with torch.no_grad():
+            relinked = self.env.add_communication_links_batched(
+                gnext.detach())
+            self.h_new.copy_(self.algo.cbf(relinked))
+
+    def _back(self, gcur, gnext, actions):
+        """Doubled-batch CBF + weighted losses + autograd.grad."""
+        algo, env = self.algo, self.env
+        p = algo.params
+        eps, alpha = p["eps"], p["alpha"]
+
+        states2 = torch.cat([gcur.states, gnext.states], dim=0)
+        ea2 = torch.cat([gcur.edge_attr, gnext.edge_attr], dim=0)
+        ei2 = torch.cat([gcur.edge_index, gcur.edge_index + self.Ntot],
+                        dim=1)
+        seg = gcur.seg_dst
+        # pad sentinel N_tot must stay past the LAST node of the doubled
+        # batch (it would otherwise alias the second half's first node)
+        pad = seg == self.Ntot
+        seg2 = torch.cat([torch.where(pad, 2 * self.Ntot, seg),
+                          seg + self.Ntot])
+        both = GraphBatch(x=self.x2_tile, pos=states2[:, :self.pd],
+                          states=states2, edge_index=ei2, edge_attr=ea2,
+                          ptr=self.ptr2)
+        both.seg_dst = seg2
+        h_both = algo.cbf(both)
+        h, h_next = h_both[:self.Ntot], h_both[self.Ntot:]
+        hv = h[:, 0]
+
+        w_node = self.w_dev.view(self.G_cap, 1).expand(
+            self.G_cap, self.N).reshape(self.Ntot)
+        cw = w_node.sum()
+
+        # identical weighted-sum forms as GCBF._iter_eager, with the pad
+        # rows carrying weight 0
+        wu = self.env.unsafe_mask(gcur).to(hv.dtype) * w_node
+        cu = wu.sum()
+        cu1 = cu.clamp(min=1)
+        any_u = (cu > 0).to(hv.dtype)
+        loss_unsafe = any_u * (torch.relu(hv + eps) * wu).sum() / cu1
+        acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
+                      + (1 - any_u))
+
+        ws = self.env.safe_mask(gcur).to(hv.dtype) * w_node
+        cs = ws.sum()
+        cs1 = cs.clamp(min=1)
+        any_s = (cs > 0).to(hv.dtype)
+        loss_safe = any_s * (torch.relu(-hv + eps) * ws).sum() / cs1
+        acc_safe = (any_s * ((hv >= 0).to(hv.dtype) * ws).sum() / cs1
+                    + (1 - any_s))
+
+        h_dot = (h_next - h) / env.dt
+        h_dot_new_link = (self.h_new - h) / env.dt
+        residue = (h_dot_new_link - h_dot).detach()
+        h_dot = residue + h_dot
+        hd = h_dot[:, 0]
+        ha = hv * alpha
+        loss_h_dot = (torch.relu(-hd - ha + eps) * w_node).sum() / cw
+        acc_h_dot = (((hd + ha) >= 0).to(hv.dtype) * w_node).sum() / cw
+
+        loss_action = (torch.square(actions).sum(dim=1) * w_node).sum() / cw
+
+        loss = (p["loss_unsafe_coef"] * loss_unsafe +
+                p["loss_safe_coef"] * loss_safe +
+                p["loss_h_dot_coef"] * loss_h_dot +
+                p["loss_action_coef"] * loss_action)
+
+        params = self._params
+        grads = torch.autograd.grad(loss, params)
+        log7 = torch.stack([
+            loss_unsafe.detach(), loss_safe.detach(), loss_h_dot.detach(),
+            loss_action.detach(), acc_unsafe.detach(), acc_safe.detach(),
+            acc_h_dot.detach()])
+        return grads, log7
+
+    # ------------------------------------------------------------ capture
+    def _fill_inputs(self, graph_list):
+        L = len(graph_list)
+        ids = [g.ring_id for g in graph_list]
+        self.idx_host[:L] = torch.tensor([i % self.CAP for i in ids],
+                                         dtype=torch.long)
+        self.idx_host[L:] = ids[0] % self.CAP
+        self.w_host[:L] = 1.0
+        self.w_host[L:] = 0.0
+        self.idx_dev.copy_(self.idx_host, non_blocking=True)
+        self.w_dev.copy_(self.w_host, non_blocking=True)
+
+    def _sample_for_warmup(self):
+        algo = self.algo
+        seg_len = 3
+        if algo.memory.size == 0:
+            return algo.buffer.sample(algo.batch_size // 5, seg_len)
+        curr = algo.buffer.sample(algo.batch_size // 10, seg_len, True)
+        prev = algo.memory.sample(
+            algo.batch_size // 5 - algo.batch_size // 10, seg_len, True)
+        return curr + prev
+
+    def _opt_tail(self, grads):
+        algo = self.algo
+        for prm, g in zip(self._params, grads):
+            prm.grad = g
+        if algo.grad_sync is not None:
+            algo.grad_sync()
+        torch.nn.utils.clip_grad_norm_(algo.cbf.parameters(), 1e-3)
+        torch.nn.utils.clip_grad_norm_(algo.actor.parameters(), 1e-3)
+        algo.optim_cbf.step()
+        algo.optim_actor.step()
+        # captured graphs (rollout AND this engine's) read the bf16
+        # mirrors by address — refresh after every weight change
+        from .nn.fused import sync_bf16_mirrors
+        sync_bf16_mirrors(algo.actor)
+        sync_bf16_mirrors(algo.cbf)
+
+    def _build(self):
+        algo = self.algo
+        self._params = [p for p in algo.cbf.parameters()
+                        if p.requires_grad] + \
+                       [p for p in algo.actor.parameters()
+                        if p.requires_grad]
+
+        # warmup: real iterations over the engine's padded shapes (warms
+        # hipBLASLt shape caches, materializes Adam state)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(self.WARMUP_ITERS):
+                self._fill_inputs(self._sample_for_warmup())
+                gcur, gnext, actions, ecount = self._front()
+                if int(ecount.cpu()[0]) > self.E_cap:
+                    raise RuntimeError("edge overflow during warmup")
+                self._middle(gnext)
+                grads, _ = self._back(gcur, gnext, actions)
+                self._opt_tail(grads)
+        torch.cuda.current_stream().wait_stream(s)
+
+        self._fill_inputs(self._sample_for_warmup())
+        self.gF = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.gF):
+            self._gcur, self._gnext, self._actions, self._ecount = \
+                self._front()
+        # capture records without executing, so the static buffers hold
+        # uninitialized pool memory here — replay once to materialize real
+        # contents before the eager middle runs on them
+        self.gF.replay()
+        self._middle(self._gnext)
+        self.gB = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.gB, pool=self.gF.pool()):
+            self._grads, self._log7 = self._back(
+                self._gcur, self._gnext, self._actions)
+
+    # -------------------------------------------------------------- iter
+    def try_iter(self, graph_list):
+        """One captured inner iteration; returns the 7-scalar log stack,
+        or None to request the exact eager fallback (no weights touched)."""
+        if len(graph_list) > self.G_cap:
+            return None
+        if any(g.ring_id is None for g in graph_list):
+            return None
+        # goal address-stability across env resets
+        if self.env._goal is not self.goal_static:
+            self.goal_static.copy_(self.env._goal)
+            self.env._goal = self.goal_static
+        self._fill_inputs(graph_list)
+        self.gF.replay()
+        if int(self._ecount.cpu()[0]) > self.E_cap:
+            return None
+        self._middle(self._gnext)
+        self.gB.replay()
+        self._opt_tail(self._grads)
+        return self._log7.clone()

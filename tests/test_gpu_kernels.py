@@ -641,3 +641,82 @@ def test_rollout_engine_soft_capacity_overflow_fallback():
     assert saw_captured
     # with capacity this tight the moving scene must overflow at least once
     assert saw_eager
+
+
+def test_update_engine_matches_eager():
+    """The hipGraph-captured update engine must produce the same losses and
+    weight updates as the eager inner iterations, from identical state."""
+    import copy
+    import random
+    from gcbf_amd.env import make_env
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.rollout import RolloutEngine
+    from gcbf_amd.trainer.utils import set_seed
+    from gcbf_amd.utils.amp import enable_bf16
+
+    set_seed(11)
+    dev = torch.device("cuda")
+    env = make_env("DubinsCar", 16, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=512)
+    enable_bf16(algo)
+    env.reset()
+    eng = RolloutEngine(env, algo)
+
+    def rollout(k):
+        for _ in range(k):
+            if eng.step(prob=0.7):
+                eng.reload()
+
+    rollout(512)
+    algo.update(512)                      # engine init + first update
+    assert algo._upd_engine is not None, "update engine must build on GPU"
+    rollout(512)
+
+    def snapshot():
+        return dict(
+            cbf=copy.deepcopy(algo.cbf.state_dict()),
+            actor=copy.deepcopy(algo.actor.state_dict()),
+            ocbf=copy.deepcopy(algo.optim_cbf.state_dict()),
+            oact=copy.deepcopy(algo.optim_actor.state_dict()),
+            buf=(list(algo.buffer.data), list(algo.buffer.safe_data),
+                 list(algo.buffer.unsafe_data), list(algo.buffer._pending)),
+            mem=(list(algo.memory.data), list(algo.memory.safe_data),
+                 list(algo.memory.unsafe_data), list(algo.memory._pending)),
+            np_state=np.random.get_state(), py_state=random.getstate())
+
+    def restore(s):
+        algo.cbf.load_state_dict(s["cbf"])
+        algo.actor.load_state_dict(s["actor"])
+        algo.optim_cbf.load_state_dict(s["ocbf"])
+        algo.optim_actor.load_state_dict(s["oact"])
+        (algo.buffer._data, algo.buffer.safe_data, algo.buffer.unsafe_data,
+         algo.buffer._pending) = [list(v) for v in s["buf"]]
+        (algo.memory._data, algo.memory.safe_data, algo.memory.unsafe_data,
+         algo.memory._pending) = [list(v) for v in s["mem"]]
+        np.random.set_state(s["np_state"])
+        random.setstate(s["py_state"])
+        from gcbf_amd.nn.fused import sync_bf16_mirrors
+        sync_bf16_mirrors(algo.cbf)
+        sync_bf16_mirrors(algo.actor)
+
+    s0 = snapshot()
+    import os
+    os.environ["GCBF_AMD_UPDATE_PROF"] = "0"
+    algo.update(1024)
+    w_eng = {k: v.clone() for k, v in algo.cbf.state_dict().items()}
+    a_eng = {k: v.clone() for k, v in algo.actor.state_dict().items()}
+
+    restore(s0)
+    saved_engine, algo._upd_engine = algo._upd_engine, None
+    algo.update(1024)
+    algo._upd_engine = saved_engine
+
+    for name, ref_sd in (("cbf", w_eng), ("actor", a_eng)):
+        sd = (algo.cbf if name == "cbf" else algo.actor).state_dict()
+        for k, v in ref_sd.items():
+            if not torch.is_floating_point(v):
+                continue
+            d = (v.float() - sd[k].float()).abs().max().item()
+            assert d < 5e-4, f"{name}.{k}: engine vs eager diff {d}"
