@@ -148,7 +148,10 @@ class GCBF(Algorithm):
             from ..update_engine import UpdateEngine
             self._upd_engine = UpdateEngine(self, env)
         except Exception as e:  # fall back to eager iterations
+            import traceback
             import warnings
+            if os.environ.get("GCBF_AMD_UPDATE_CAPTURE_DEBUG") == "1":
+                traceback.print_exc()
             warnings.warn(f"update capture unavailable ({e}); eager updates")
             self._upd_engine = None
 
