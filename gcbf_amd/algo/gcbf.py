@@ -141,7 +141,11 @@ class GCBF(Algorithm):
             return
         env = self._env
         data = env.data
-        if (data is None or data.agent_mask is not None
+        # v1 requires every node to be an agent (no obstacle points): an
+        # all-agent graph makes the boolean agent mask a mathematical no-op,
+        # so the captured pipeline can drop it (boolean indexing is not
+        # capturable)
+        if (data is None or data.num_nodes != env.num_agents
                 or getattr(env, "_max_neighbors", None) is not None):
             return
         try:

@@ -60,7 +60,7 @@ class UpdateEngine:
         self.device = env.device
 
         data = env.data
-        assert data is not None and data.agent_mask is None
+        assert data is not None and data.num_nodes == env.num_agents
         self.N = data.num_nodes            # nodes per graph (= agents here)
         self.S = env.state_dim
         self.pd = 3 if env.state_dim == 6 else 2
