@@ -1,19 +1,119 @@
-import os, torch
+"""Diagnostic: captured update engine vs eager updates — per-iteration
+losses, final weight diffs, and wall time.  Run on a GPU box:
+
+    PYTHONPATH=. python tools/upde_repro.py
+"""
+import copy
+import os
+import random
+import time
+
+import numpy as np
+import torch
+
 os.environ["GCBF_AMD_UPDATE_CAPTURE_DEBUG"] = "1"
+
 from gcbf_amd.env import make_env
 from gcbf_amd.algo import make_algo
 from gcbf_amd.rollout import RolloutEngine
 from gcbf_amd.trainer.utils import set_seed
 from gcbf_amd.utils.amp import enable_bf16
+from gcbf_amd.nn.fused import sync_bf16_mirrors
+
 set_seed(11)
 dev = torch.device("cuda")
-env = make_env("DubinsCar", 16, dev); env.train()
-algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim, env.action_dim, dev, batch_size=512)
+env = make_env("DubinsCar", 16, dev)
+env.train()
+algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                 env.action_dim, dev, batch_size=512)
 enable_bf16(algo)
 env.reset()
 eng = RolloutEngine(env, algo)
-for _ in range(512):
-    if eng.step(prob=0.7):
-        eng.reload()
+
+
+def rollout(k):
+    for _ in range(k):
+        if eng.step(prob=0.7):
+            eng.reload()
+
+
+rollout(512)
 algo.update(512)
-print("engine:", algo._upd_engine)
+print("engine:", type(algo._upd_engine).__name__ if algo._upd_engine
+      else None)
+if algo._upd_engine:
+    print("E_cap:", algo._upd_engine.E_cap, "G_cap:", algo._upd_engine.G_cap)
+rollout(512)
+
+
+def snapshot():
+    return dict(
+        cbf=copy.deepcopy(algo.cbf.state_dict()),
+        actor=copy.deepcopy(algo.actor.state_dict()),
+        ocbf=copy.deepcopy(algo.optim_cbf.state_dict()),
+        oact=copy.deepcopy(algo.optim_actor.state_dict()),
+        buf=(list(algo.buffer.data), list(algo.buffer.safe_data),
+             list(algo.buffer.unsafe_data), list(algo.buffer._pending)),
+        mem=(list(algo.memory.data), list(algo.memory.safe_data),
+             list(algo.memory.unsafe_data), list(algo.memory._pending)),
+        np_state=np.random.get_state(), py_state=random.getstate())
+
+
+def restore(s):
+    algo.cbf.load_state_dict(s["cbf"])
+    algo.actor.load_state_dict(s["actor"])
+    algo.optim_cbf.load_state_dict(s["ocbf"])
+    algo.optim_actor.load_state_dict(s["oact"])
+    (algo.buffer._data, algo.buffer.safe_data, algo.buffer.unsafe_data,
+     algo.buffer._pending) = [list(v) for v in s["buf"]]
+    (algo.memory._data, algo.memory.safe_data, algo.memory.unsafe_data,
+     algo.memory._pending) = [list(v) for v in s["mem"]]
+    np.random.set_state(s["np_state"])
+    random.setstate(s["py_state"])
+    sync_bf16_mirrors(algo.cbf)
+    sync_bf16_mirrors(algo.actor)
+
+
+captured_logs = {}
+orig_tail = algo._update_tail
+
+
+def spy_tail(step, writer, logs, inner_iter):
+    captured_logs["logs"] = [t.clone() for t in logs]
+    return orig_tail(step, writer, logs, inner_iter)
+
+
+algo._update_tail = spy_tail
+
+s0 = snapshot()
+torch.cuda.synchronize(); t0 = time.perf_counter()
+algo.update(1024)
+torch.cuda.synchronize()
+t_eng = time.perf_counter() - t0
+eng_logs = captured_logs["logs"]
+w_eng = {f"cbf.{k}": v.clone() for k, v in algo.cbf.state_dict().items()}
+w_eng.update({f"actor.{k}": v.clone()
+              for k, v in algo.actor.state_dict().items()})
+
+restore(s0)
+saved, algo._upd_engine = algo._upd_engine, None
+torch.cuda.synchronize(); t0 = time.perf_counter()
+algo.update(1024)
+torch.cuda.synchronize()
+t_eag = time.perf_counter() - t0
+algo._upd_engine = saved
+eag_logs = captured_logs["logs"]
+
+print(f"time: engine {t_eng*1e3:.1f} ms  eager {t_eag*1e3:.1f} ms")
+for i, (a, b) in enumerate(zip(eng_logs, eag_logs)):
+    d = (a - b).abs()
+    print(f"iter {i}: engine {a.tolist()}")
+    print(f"         eager {b.tolist()}  maxdiff {d.max().item():.3e}")
+diffs = []
+w_now = {f"cbf.{k}": v for k, v in algo.cbf.state_dict().items()}
+w_now.update({f"actor.{k}": v for k, v in algo.actor.state_dict().items()})
+for k, v in w_eng.items():
+    if torch.is_floating_point(v):
+        diffs.append(((v.float() - w_now[k].float()).abs().max().item(), k))
+diffs.sort(reverse=True)
+print("top weight diffs:", [(f"{d:.2e}", k) for d, k in diffs[:6]])
