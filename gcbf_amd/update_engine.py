@@ -292,8 +292,14 @@ class UpdateEngine:
         torch.cuda.current_stream().wait_stream(s)
 
         self._fill_inputs(self._sample_for_warmup())
+        # ONE capture stream for BOTH graphs: backward kernels are launched
+        # on the stream their forward op was captured on, so capturing the
+        # forward (graph F) and its backward (graph B) on different streams
+        # would leak the cross-graph backward segments out of the capture
+        # (observed as stale gradients on replay)
+        cap_stream = torch.cuda.Stream()
         self.gF = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.gF):
+        with torch.cuda.graph(self.gF, stream=cap_stream):
             self._gcur, self._gnext, self._actions, self._ecount = \
                 self._front()
         # capture records without executing, so the static buffers hold
@@ -302,7 +308,8 @@ class UpdateEngine:
         self.gF.replay()
         self._middle(self._gnext)
         self.gB = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.gB, pool=self.gF.pool()):
+        with torch.cuda.graph(self.gB, pool=self.gF.pool(),
+                              stream=cap_stream):
             self._grads, self._log7 = self._back(
                 self._gcur, self._gnext, self._actions)
 
