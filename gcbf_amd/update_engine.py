@@ -1,20 +1,31 @@
 """hipGraph-captured training update engine.
 
 At small scenes (the headline DubinsCar n=16 config) one eager inner
-update iteration costs ~15 ms of which most is kernel-launch/Python gaps:
-~300 tiny graphs are re-batched in Python, then ~200 small kernels run for
-the four GNN forwards, the losses and the backward.  This engine replays
-the whole iteration as two hipGraphs over fixed-capacity buffers:
+update iteration costs ~15-30 ms of which most is kernel-launch/Python
+gaps: ~300 tiny graphs are re-batched in Python, then ~200 small kernels
+run for the four GNN forwards, the losses and the backward.  This engine
+replays the whole iteration as two hipGraphs over fixed-capacity buffers:
 
-    graph F:  gather sampled graphs from a device ring (index_select)
-              → padded batched radius-graph build (fixed E_cap)
-              → actor forward → env.forward_graph (next states/attrs)
-    eager  :  exact re-link build + CBF forward of the re-linked next
-              graph (the ḣ residue trick needs EXACT edges — this stays
-              out of capture so overflow cannot corrupt gradients)
-    graph B:  doubled-batch CBF forward (h and h_next under one σ)
-              → weighted-sum losses → torch.autograd.grad
-    eager  :  grad all-reduce (DP) + clip + Adam + bf16 mirror refresh
+    graph FRONT (no_grad): gather sampled graphs from a device ring
+        (index_select) → padded batched radius-graph build (fixed E_cap)
+        → actor forward → env.forward_graph → padded RE-LINK build of the
+        next graph → CBF forward of the re-linked graph (the ḣ residue)
+        → publish both edge counts
+    one 8-byte host read of the two edge counts (overflow → exact eager
+        fallback; FRONT touches no weights, so fallback is always safe)
+    graph BACK (self-contained fwd+bwd): actor forward → forward_graph →
+        doubled-batch CBF (h and h_next under one spectral-norm σ) →
+        weighted-sum losses → torch.autograd.grad
+    eager tail: grad all-reduce (DP) + clip + Adam + bf16 mirror refresh
+
+Why BACK recomputes the actor: autograd must not cross capture
+boundaries (backward kernels are launched on the stream their forward was
+captured on, so a backward recorded in a later graph silently leaks out of
+the capture — observed as stale gradients).  BACK is therefore captured
+exactly like ``torch.cuda.make_graphed_callables`` captures a callable:
+forward and backward in one recording.  FRONT's actor/next-state kernels
+are the same recorded kernel sequence on the same buffers, so its re-link
+residue is bitwise consistent with BACK's differentiable path.
 
 Design points:
 * The replay buffer's graphs are mirrored into a device ring of states +
@@ -24,18 +35,15 @@ Design points:
   them at rollout time), so the ring fully reconstructs any sampled graph.
 * Static shapes come from padding: the graph count is padded to G_cap
   (3 · batch_size/5, the sampler's maximum) with weight-0 copies of a real
-  graph, and the edge buffers hold E_cap entries with the same
-  sentinel-segment scheme as the rollout engine.  Every loss/accuracy mean
-  becomes a weighted sum, which is exactly equal on the real rows.
-* Gradients come from ``torch.autograd.grad`` captured in graph B; the
-  resulting tensors are assigned to ``param.grad`` once, so the eager
-  optimizer tail (and the DP flat-buffer all-reduce) read refreshed
-  contents each replay.  Optimizer, clipping and communication stay
-  eager — no capturable-optimizer or captured-collective requirements,
-  and the engine composes with data parallelism unchanged.
-* Overflow of E_cap is detected from the published edge count AFTER
-  graph F (which touches no weights); the iteration falls back to the
-  exact eager path (``GCBF._iter_eager``).
+  graph, and edge buffers hold E_cap entries with the rollout engine's
+  sentinel-segment scheme.  Every loss/accuracy mean becomes a weighted
+  sum, exactly equal on the real rows.
+* Gradients come from ``torch.autograd.grad`` captured in BACK; the
+  resulting tensors are (re)assigned to ``param.grad`` every iteration, so
+  the eager optimizer tail and the DP flat-buffer all-reduce read
+  refreshed contents each replay.  Optimizer, clipping and communication
+  stay eager — no capturable-optimizer or captured-collective
+  requirements, and the engine composes with data parallelism unchanged.
 
 Reference behavior preserved (gcbf/algo/gcbf.py:140-230): sampling,
 balanced replay, the ḣ re-link residue trick, loss forms and coefficients,
@@ -77,7 +85,7 @@ class UpdateEngine:
         # edge capacity: a sampled batch of G_cap graphs concentrates hard
         # around G_cap·mean_e edges (~300 draws), so a 1.35x headroom keeps
         # overflow (exact eager fallback) rare while padded GEMM rows — pure
-        # wasted FLOPs — stay modest.  4x padding measured 3x slower updates.
+        # wasted FLOPs — stay modest (4x padding measured 3x slower updates)
         mean_e = max(1.0, sum(g.num_edges for g in algo.buffer.data)
                      / algo.buffer.size)
         full = self.G_cap * self.N * (self.N - 1)
@@ -106,8 +114,6 @@ class UpdateEngine:
         self.idx_dev = torch.zeros(self.G_cap, dtype=torch.long, device=dev)
         self.w_dev = torch.zeros(self.G_cap, device=dev)
 
-        self.h_new = torch.zeros(self.Ntot, 1, device=dev)
-
         # goal must be address-stable: forward_graph reads env._goal (the
         # reference's replayed-graph u_ref uses the CURRENT goal)
         self.goal_static = env._goal.clone().contiguous()
@@ -131,41 +137,55 @@ class UpdateEngine:
         self.next_id += 1
 
     # ------------------------------------------------------------- bodies
-    def _front(self):
-        """Gather + padded build + actor forward + env.forward_graph."""
+    def _gather(self):
+        nodes = self.ring_states.index_select(
+            0, self.idx_dev).reshape(self.Ntot, self.S)
+        uref = self.ring_uref.index_select(
+            0, self.idx_dev).reshape(self.Ntot, self.ad)
+        return nodes, uref
+
+    def _build_cur(self, nodes, uref):
         env = self.env
-        with torch.no_grad():
-            nodes = self.ring_states.index_select(
-                0, self.idx_dev).reshape(self.Ntot, self.S)
-            uref = self.ring_uref.index_select(
-                0, self.idx_dev).reshape(self.Ntot, self.ad)
-            ei, seg, ea, ecount = self._ext.build_graph_padded(
-                nodes[:, :self.pd].contiguous(), nodes, self.G_cap, self.N,
-                env.params["comm_radius"], -1, env._attr_kind, self.ed,
-                self.E_cap)
+        ei, seg, ea, ecount = self._ext.build_graph_padded(
+            nodes[:, :self.pd].contiguous(), nodes, self.G_cap, self.N,
+            env.params["comm_radius"], -1, env._attr_kind, self.ed,
+            self.E_cap)
         gcur = GraphBatch(x=self.x_tile, pos=nodes[:, :self.pd],
                           states=nodes, edge_index=ei, edge_attr=ea,
                           u_ref=uref, ptr=self.ptr)
         gcur.seg_dst = seg
-        actions = self.algo.actor(gcur)
-        gnext = env.forward_graph(gcur, actions)
-        gnext.seg_dst = seg
-        return gcur, gnext, actions, ecount
+        return gcur, ecount
 
-    def _middle(self, gnext):
-        """EXACT re-link + CBF forward of the re-linked next graph (eager;
-        the residue needs exact edges, and overflow must not be able to
-        corrupt gradients)."""
+    def _front(self):
+        """no_grad: gather + build + actor + next states + padded re-link
+        + CBF of the re-linked graph (the ḣ residue value)."""
+        env, algo = self.env, self.algo
         with torch.no_grad():
-            relinked = self.env.add_communication_links_batched(
-                gnext.detach())
-            self.h_new.copy_(self.algo.cbf(relinked))
+            nodes, uref = self._gather()
+            gcur, ecount = self._build_cur(nodes, uref)
+            actions = algo.actor(gcur)
+            gnext = env.forward_graph(gcur, actions)
+            ns = gnext.states
+            ei2, seg2, ea2, ecount2 = self._ext.build_graph_padded(
+                ns[:, :self.pd].contiguous(), ns.contiguous(), self.G_cap,
+                self.N, env.params["comm_radius"], -1, env._attr_kind,
+                self.ed, self.E_cap)
+            grel = GraphBatch(x=self.x_tile, pos=ns[:, :self.pd], states=ns,
+                              edge_index=ei2, edge_attr=ea2, ptr=self.ptr)
+            grel.seg_dst = seg2
+            h_new = algo.cbf(grel)
+            ecounts = torch.stack([ecount[0], ecount2[0]])
+        return gcur, h_new, ecounts
 
-    def _back(self, gcur, gnext, actions):
-        """Doubled-batch CBF + weighted losses + autograd.grad."""
+    def _back(self, gcur, h_new):
+        """Self-contained fwd+bwd: actor → forward_graph → doubled CBF →
+        weighted losses → autograd.grad (captured in ONE graph)."""
         algo, env = self.algo, self.env
         p = algo.params
         eps, alpha = p["eps"], p["alpha"]
+
+        actions = algo.actor(gcur)
+        gnext = env.forward_graph(gcur, actions)
 
         states2 = torch.cat([gcur.states, gnext.states], dim=0)
         ea2 = torch.cat([gcur.edge_attr, gnext.edge_attr], dim=0)
@@ -191,7 +211,7 @@ class UpdateEngine:
 
         # identical weighted-sum forms as GCBF._iter_eager, with the pad
         # rows carrying weight 0
-        wu = self.env.unsafe_mask(gcur).to(hv.dtype) * w_node
+        wu = env.unsafe_mask(gcur).to(hv.dtype) * w_node
         cu = wu.sum()
         cu1 = cu.clamp(min=1)
         any_u = (cu > 0).to(hv.dtype)
@@ -199,7 +219,7 @@ class UpdateEngine:
         acc_unsafe = (any_u * ((hv < 0).to(hv.dtype) * wu).sum() / cu1
                       + (1 - any_u))
 
-        ws = self.env.safe_mask(gcur).to(hv.dtype) * w_node
+        ws = env.safe_mask(gcur).to(hv.dtype) * w_node
         cs = ws.sum()
         cs1 = cs.clamp(min=1)
         any_s = (cs > 0).to(hv.dtype)
@@ -208,7 +228,7 @@ class UpdateEngine:
                     + (1 - any_s))
 
         h_dot = (h_next - h) / env.dt
-        h_dot_new_link = (self.h_new - h) / env.dt
+        h_dot_new_link = (h_new - h) / env.dt
         residue = (h_dot_new_link - h_dot).detach()
         h_dot = residue + h_dot
         hd = h_dot[:, 0]
@@ -223,8 +243,7 @@ class UpdateEngine:
                 p["loss_h_dot_coef"] * loss_h_dot +
                 p["loss_action_coef"] * loss_action)
 
-        params = self._params
-        grads = torch.autograd.grad(loss, params)
+        grads = torch.autograd.grad(loss, self._params)
         log7 = torch.stack([
             loss_unsafe.detach(), loss_safe.detach(), loss_h_dot.detach(),
             loss_action.detach(), acc_unsafe.detach(), acc_safe.detach(),
@@ -255,6 +274,9 @@ class UpdateEngine:
 
     def _opt_tail(self, grads):
         algo = self.algo
+        # reassign every iteration: an interleaved eager fallback iteration
+        # (zero_grad(set_to_none=True) + backward) would otherwise leave
+        # param.grad pointing at its own tensors, not the captured ones
         for prm, g in zip(self._params, grads):
             prm.grad = g
         if algo.grad_sync is not None:
@@ -283,35 +305,23 @@ class UpdateEngine:
         with torch.cuda.stream(s):
             for _ in range(self.WARMUP_ITERS):
                 self._fill_inputs(self._sample_for_warmup())
-                gcur, gnext, actions, ecount = self._front()
-                if int(ecount.cpu()[0]) > self.E_cap:
+                gcur, h_new, ecounts = self._front()
+                if int(ecounts.max().cpu()) > self.E_cap:
                     raise RuntimeError("edge overflow during warmup")
-                self._middle(gnext)
-                grads, _ = self._back(gcur, gnext, actions)
+                grads, _ = self._back(gcur, h_new)
                 self._opt_tail(grads)
         torch.cuda.current_stream().wait_stream(s)
 
         self._fill_inputs(self._sample_for_warmup())
-        # ONE capture stream for BOTH graphs: backward kernels are launched
-        # on the stream their forward op was captured on, so capturing the
-        # forward (graph F) and its backward (graph B) on different streams
-        # would leak the cross-graph backward segments out of the capture
-        # (observed as stale gradients on replay)
-        cap_stream = torch.cuda.Stream()
-        self.gF = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.gF, stream=cap_stream):
-            self._gcur, self._gnext, self._actions, self._ecount = \
-                self._front()
-        # capture records without executing, so the static buffers hold
-        # uninitialized pool memory here — replay once to materialize real
-        # contents before the eager middle runs on them
-        self.gF.replay()
-        self._middle(self._gnext)
-        self.gB = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.gB, pool=self.gF.pool(),
-                              stream=cap_stream):
-            self._grads, self._log7 = self._back(
-                self._gcur, self._gnext, self._actions)
+        self.gFront = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.gFront):
+            self._gcur, self._h_new, self._ecounts = self._front()
+        # capture records without executing — replay once so the static
+        # buffers hold real contents before BACK's capture
+        self.gFront.replay()
+        self.gBack = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.gBack, pool=self.gFront.pool()):
+            self._grads, self._log7 = self._back(self._gcur, self._h_new)
 
     # -------------------------------------------------------------- iter
     def try_iter(self, graph_list):
@@ -326,10 +336,9 @@ class UpdateEngine:
             self.goal_static.copy_(self.env._goal)
             self.env._goal = self.goal_static
         self._fill_inputs(graph_list)
-        self.gF.replay()
-        if int(self._ecount.cpu()[0]) > self.E_cap:
+        self.gFront.replay()
+        if int(self._ecounts.max().cpu()) > self.E_cap:
             return None
-        self._middle(self._gnext)
-        self.gB.replay()
+        self.gBack.replay()
         self._opt_tail(self._grads)
         return self._log7.clone()
