@@ -1,24 +1,19 @@
-"""Diagnostic: captured update engine vs eager updates — per-iteration
-losses, final weight diffs, and wall time.  Run on a GPU box:
+"""Diagnostic v2: captured update engine — gradient and staleness probes.
 
     PYTHONPATH=. python tools/upde_repro.py
 """
-import copy
 import os
-import random
-import time
 
-import numpy as np
 import torch
 
 os.environ["GCBF_AMD_UPDATE_CAPTURE_DEBUG"] = "1"
 
 from gcbf_amd.env import make_env
 from gcbf_amd.algo import make_algo
+from gcbf_amd.graph import GraphBatch
 from gcbf_amd.rollout import RolloutEngine
 from gcbf_amd.trainer.utils import set_seed
 from gcbf_amd.utils.amp import enable_bf16
-from gcbf_amd.nn.fused import sync_bf16_mirrors
 
 set_seed(11)
 dev = torch.device("cuda")
@@ -29,91 +24,97 @@ algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
 enable_bf16(algo)
 env.reset()
 eng = RolloutEngine(env, algo)
-
-
-def rollout(k):
-    for _ in range(k):
-        if eng.step(prob=0.7):
-            eng.reload()
-
-
-rollout(512)
+for _ in range(512):
+    if eng.step(prob=0.7):
+        eng.reload()
 algo.update(512)
-print("engine:", type(algo._upd_engine).__name__ if algo._upd_engine
-      else None)
-if algo._upd_engine:
-    print("E_cap:", algo._upd_engine.E_cap, "G_cap:", algo._upd_engine.G_cap)
-rollout(512)
+e = algo._upd_engine
+print("engine:", type(e).__name__ if e else None)
+assert e is not None
+for _ in range(512):
+    if eng.step(prob=0.7):
+        eng.reload()
+
+names = [f"cbf.{n}" for n, p in algo.cbf.named_parameters()
+         if p.requires_grad] + \
+        [f"actor.{n}" for n, p in algo.actor.named_parameters()
+         if p.requires_grad]
 
 
-def snapshot():
-    return dict(
-        cbf=copy.deepcopy(algo.cbf.state_dict()),
-        actor=copy.deepcopy(algo.actor.state_dict()),
-        ocbf=copy.deepcopy(algo.optim_cbf.state_dict()),
-        oact=copy.deepcopy(algo.optim_actor.state_dict()),
-        buf=(list(algo.buffer.data), list(algo.buffer.safe_data),
-             list(algo.buffer.unsafe_data), list(algo.buffer._pending)),
-        mem=(list(algo.memory.data), list(algo.memory.safe_data),
-             list(algo.memory.unsafe_data), list(algo.memory._pending)),
-        np_state=np.random.get_state(), py_state=random.getstate())
+def eager_loss(gl):
+    """Replicate GCBF._iter_eager's loss (no optimizer step)."""
+    p = algo.params
+    eps, alpha = p["eps"], p["alpha"]
+    graphs = GraphBatch.from_list(gl)
+    actions = algo.actor(graphs)
+    graphs_next = env.forward_graph(graphs, actions)
+    both = GraphBatch.from_list([graphs, graphs_next])
+    h_both = algo.cbf(both)
+    n_ag = h_both.shape[0] // 2
+    h, h_next = h_both[:n_ag], h_both[n_ag:]
+    hv = h[:, 0]
+    um = env.unsafe_mask(graphs).to(hv.dtype)
+    cu1 = um.sum().clamp(min=1)
+    any_u = (um.sum() > 0).to(hv.dtype)
+    loss_unsafe = any_u * (torch.relu(hv + eps) * um).sum() / cu1
+    sm = env.safe_mask(graphs).to(hv.dtype)
+    cs1 = sm.sum().clamp(min=1)
+    any_s = (sm.sum() > 0).to(hv.dtype)
+    loss_safe = any_s * (torch.relu(-hv + eps) * sm).sum() / cs1
+    with torch.no_grad():
+        relinked = env.add_communication_links_batched(graphs_next.detach())
+        h_new = algo.cbf(relinked)
+    h_dot = (h_next - h) / env.dt
+    residue = ((h_new - h) / env.dt - h_dot).detach()
+    h_dot = residue + h_dot
+    loss_h_dot = torch.mean(torch.relu(-h_dot - alpha * h + eps))
+    loss_action = torch.mean(torch.square(actions).sum(dim=1))
+    loss = (p["loss_unsafe_coef"] * loss_unsafe +
+            p["loss_safe_coef"] * loss_safe +
+            p["loss_h_dot_coef"] * loss_h_dot +
+            p["loss_action_coef"] * loss_action)
+    return loss, [loss_unsafe, loss_safe, loss_h_dot, loss_action]
 
 
-def restore(s):
-    algo.cbf.load_state_dict(s["cbf"])
-    algo.actor.load_state_dict(s["actor"])
-    algo.optim_cbf.load_state_dict(s["ocbf"])
-    algo.optim_actor.load_state_dict(s["oact"])
-    (algo.buffer._data, algo.buffer.safe_data, algo.buffer.unsafe_data,
-     algo.buffer._pending) = [list(v) for v in s["buf"]]
-    (algo.memory._data, algo.memory.safe_data, algo.memory.unsafe_data,
-     algo.memory._pending) = [list(v) for v in s["mem"]]
-    np.random.set_state(s["np_state"])
-    random.setstate(s["py_state"])
-    sync_bf16_mirrors(algo.cbf)
-    sync_bf16_mirrors(algo.actor)
+gl = e._sample_for_warmup()
 
+# ---- probe 1: replayed grads vs eager grads on the SAME batch
+e._fill_inputs(gl)
+e.gFront.replay()
+assert int(e._ecounts.max().cpu()) <= e.E_cap
+e.gBack.replay()
+g_cap = [t.clone() for t in e._grads]
+log_cap = e._log7.clone()
 
-captured_logs = {}
-orig_tail = algo._update_tail
+loss, parts = eager_loss(gl)
+g_eag = torch.autograd.grad(loss, e._params)
+print("log7 cap :", [round(float(x), 5) for x in log_cap[:4]])
+print("loss eag :", [round(float(x), 5) for x in parts])
+rows = []
+for n, a, b in zip(names, g_cap, g_eag):
+    an, bn = a.float().norm().item(), b.float().norm().item()
+    cos = torch.nn.functional.cosine_similarity(
+        a.float().flatten(), b.float().flatten(), dim=0).item()
+    rows.append((cos, n, an, bn))
+rows.sort()
+print("worst grad cosines:")
+for cos, n, an, bn in rows[:8]:
+    print(f"  {cos:+.4f} {n}  |cap|={an:.3e} |eag|={bn:.3e}")
+print(f"median cosine: {sorted(r[0] for r in rows)[len(rows)//2]:+.4f}")
 
+# ---- probe 2: does the captured forward see weight changes?
+e._opt_tail(e._grads)
+e._fill_inputs(gl)
+e.gFront.replay()
+e.gBack.replay()
+log_cap2 = e._log7.clone()
+loss2, parts2 = eager_loss(gl)
+print("post-step cap:", [round(float(x), 5) for x in log_cap2[:4]])
+print("post-step eag:", [round(float(x), 5) for x in parts2])
 
-def spy_tail(step, writer, logs, inner_iter):
-    captured_logs["logs"] = [t.clone() for t in logs]
-    return orig_tail(step, writer, logs, inner_iter)
-
-
-algo._update_tail = spy_tail
-
-s0 = snapshot()
-torch.cuda.synchronize(); t0 = time.perf_counter()
-algo.update(1024)
-torch.cuda.synchronize()
-t_eng = time.perf_counter() - t0
-eng_logs = captured_logs["logs"]
-w_eng = {f"cbf.{k}": v.clone() for k, v in algo.cbf.state_dict().items()}
-w_eng.update({f"actor.{k}": v.clone()
-              for k, v in algo.actor.state_dict().items()})
-
-restore(s0)
-saved, algo._upd_engine = algo._upd_engine, None
-torch.cuda.synchronize(); t0 = time.perf_counter()
-algo.update(1024)
-torch.cuda.synchronize()
-t_eag = time.perf_counter() - t0
-algo._upd_engine = saved
-eag_logs = captured_logs["logs"]
-
-print(f"time: engine {t_eng*1e3:.1f} ms  eager {t_eag*1e3:.1f} ms")
-for i, (a, b) in enumerate(zip(eng_logs, eag_logs)):
-    d = (a - b).abs()
-    print(f"iter {i}: engine {a.tolist()}")
-    print(f"         eager {b.tolist()}  maxdiff {d.max().item():.3e}")
-diffs = []
-w_now = {f"cbf.{k}": v for k, v in algo.cbf.state_dict().items()}
-w_now.update({f"actor.{k}": v for k, v in algo.actor.state_dict().items()})
-for k, v in w_eng.items():
-    if torch.is_floating_point(v):
-        diffs.append(((v.float() - w_now[k].float()).abs().max().item(), k))
-diffs.sort(reverse=True)
-print("top weight diffs:", [(f"{d:.2e}", k) for d, k in diffs[:6]])
+# ---- probe 3: replay determinism (same weights, same inputs)
+e.gFront.replay()
+e.gBack.replay()
+log_cap3 = e._log7.clone()
+print("replay determinism maxdiff:",
+      (log_cap3 - log_cap2).abs().max().item())
