@@ -13,17 +13,22 @@ replays the whole iteration as two hipGraphs over fixed-capacity buffers:
         → publish both edge counts
     one 8-byte host read of the two edge counts (overflow → exact eager
         fallback; FRONT touches no weights, so fallback is always safe)
-    graph BACK (self-contained fwd+bwd): actor forward → forward_graph →
-        doubled-batch CBF (h and h_next under one spectral-norm σ) →
-        weighted-sum losses → torch.autograd.grad
+    graph BACK (via torch.cuda.make_graphed_callables): actor forward →
+        forward_graph → doubled-batch CBF (h and h_next under one
+        spectral-norm σ) → weighted-sum losses; ``loss.backward()``
+        replays its captured backward graph
     eager tail: grad all-reduce (DP) + clip + Adam + bf16 mirror refresh
 
 Why BACK recomputes the actor: autograd must not cross capture
 boundaries (backward kernels are launched on the stream their forward was
 captured on, so a backward recorded in a later graph silently leaks out of
-the capture — observed as stale gradients).  BACK is therefore captured
-exactly like ``torch.cuda.make_graphed_callables`` captures a callable:
-forward and backward in one recording.  FRONT's actor/next-state kernels
+the capture — observed as stale gradients).  Hand-capturing
+``torch.autograd.grad`` in one self-contained graph ALSO failed subtly:
+the returned grad tensors were pool-allocated over freed forward
+temporaries and bias gradients came back aliased (one grad had norm
+exactly 1.0 — a spectral-norm power-iteration temporary).
+``make_graphed_callables`` is torch's supported fwd+bwd capture and owns
+its static input/output/grad surfaces.  FRONT's actor/next-state kernels
 are the same recorded kernel sequence on the same buffers, so its re-link
 residue is bitwise consistent with BACK's differentiable path.
 
@@ -38,12 +43,12 @@ Design points:
   graph, and edge buffers hold E_cap entries with the rollout engine's
   sentinel-segment scheme.  Every loss/accuracy mean becomes a weighted
   sum, exactly equal on the real rows.
-* Gradients come from ``torch.autograd.grad`` captured in BACK; the
-  resulting tensors are (re)assigned to ``param.grad`` every iteration, so
-  the eager optimizer tail and the DP flat-buffer all-reduce read
-  refreshed contents each replay.  Optimizer, clipping and communication
-  stay eager — no capturable-optimizer or captured-collective
-  requirements, and the engine composes with data parallelism unchanged.
+* ``loss.backward()`` on the graphed callable replays the captured
+  backward and routes the static parameter grads through eager
+  AccumulateGrad into ``param.grad``.  Optimizer, clipping and
+  communication stay eager — no capturable-optimizer or
+  captured-collective requirements, and the engine composes with data
+  parallelism unchanged.
 
 Reference behavior preserved (gcbf/algo/gcbf.py:140-230): sampling,
 balanced replay, the ḣ re-link residue trick, loss forms and coefficients,
@@ -175,15 +180,22 @@ class UpdateEngine:
             grel.seg_dst = seg2
             h_new = algo.cbf(grel)
             ecounts = torch.stack([ecount[0], ecount2[0]])
-        return gcur, h_new, ecounts
+        return (gcur.states, gcur.u_ref, gcur.edge_index, gcur.edge_attr,
+                gcur.seg_dst, h_new, ecounts)
 
-    def _back(self, gcur, h_new):
-        """Self-contained fwd+bwd: actor → forward_graph → doubled CBF →
-        weighted losses → autograd.grad (captured in ONE graph)."""
+    def _back(self, nodes, uref, ei, ea, seg, h_new, w_dev):
+        """Differentiable stage: actor → forward_graph → doubled CBF →
+        weighted losses.  Captured via torch.cuda.make_graphed_callables
+        (hand-capturing the backward measured corrupted bias gradients —
+        grad buffers aliased freed forward temporaries in the shared pool)."""
         algo, env = self.algo, self.env
         p = algo.params
         eps, alpha = p["eps"], p["alpha"]
 
+        gcur = GraphBatch(x=self.x_tile, pos=nodes[:, :self.pd],
+                          states=nodes, edge_index=ei, edge_attr=ea,
+                          u_ref=uref, ptr=self.ptr)
+        gcur.seg_dst = seg
         actions = algo.actor(gcur)
         gnext = env.forward_graph(gcur, actions)
 
@@ -191,7 +203,6 @@ class UpdateEngine:
         ea2 = torch.cat([gcur.edge_attr, gnext.edge_attr], dim=0)
         ei2 = torch.cat([gcur.edge_index, gcur.edge_index + self.Ntot],
                         dim=1)
-        seg = gcur.seg_dst
         # pad sentinel N_tot must stay past the LAST node of the doubled
         # batch (it would otherwise alias the second half's first node)
         pad = seg == self.Ntot
@@ -205,7 +216,7 @@ class UpdateEngine:
         h, h_next = h_both[:self.Ntot], h_both[self.Ntot:]
         hv = h[:, 0]
 
-        w_node = self.w_dev.view(self.G_cap, 1).expand(
+        w_node = w_dev.view(self.G_cap, 1).expand(
             self.G_cap, self.N).reshape(self.Ntot)
         cw = w_node.sum()
 
@@ -243,12 +254,11 @@ class UpdateEngine:
                 p["loss_h_dot_coef"] * loss_h_dot +
                 p["loss_action_coef"] * loss_action)
 
-        grads = torch.autograd.grad(loss, self._params)
         log7 = torch.stack([
             loss_unsafe.detach(), loss_safe.detach(), loss_h_dot.detach(),
             loss_action.detach(), acc_unsafe.detach(), acc_safe.detach(),
             acc_h_dot.detach()])
-        return grads, log7
+        return loss, log7
 
     # ------------------------------------------------------------ capture
     def _fill_inputs(self, graph_list):
@@ -272,13 +282,8 @@ class UpdateEngine:
             algo.batch_size // 5 - algo.batch_size // 10, seg_len, True)
         return curr + prev
 
-    def _opt_tail(self, grads):
+    def _opt_tail(self):
         algo = self.algo
-        # reassign every iteration: an interleaved eager fallback iteration
-        # (zero_grad(set_to_none=True) + backward) would otherwise leave
-        # param.grad pointing at its own tensors, not the captured ones
-        for prm, g in zip(self._params, grads):
-            prm.grad = g
         if algo.grad_sync is not None:
             algo.grad_sync()
         torch.nn.utils.clip_grad_norm_(algo.cbf.parameters(), 1e-3)
@@ -291,37 +296,44 @@ class UpdateEngine:
         sync_bf16_mirrors(algo.actor)
         sync_bf16_mirrors(algo.cbf)
 
+    def _zero_grads(self):
+        self.algo.optim_cbf.zero_grad(set_to_none=True)
+        self.algo.optim_actor.zero_grad(set_to_none=True)
+
     def _build(self):
         algo = self.algo
-        self._params = [p for p in algo.cbf.parameters()
-                        if p.requires_grad] + \
-                       [p for p in algo.actor.parameters()
-                        if p.requires_grad]
+        self._back_mod = _BackCallable(self)
 
         # warmup: real iterations over the engine's padded shapes (warms
-        # hipBLASLt shape caches, materializes Adam state)
+        # hipBLASLt shape caches, materializes Adam state + bf16 mirrors)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(self.WARMUP_ITERS):
                 self._fill_inputs(self._sample_for_warmup())
-                gcur, h_new, ecounts = self._front()
-                if int(ecounts.max().cpu()) > self.E_cap:
+                front = self._front()
+                if int(front[6].max().cpu()) > self.E_cap:
                     raise RuntimeError("edge overflow during warmup")
-                grads, _ = self._back(gcur, h_new)
-                self._opt_tail(grads)
+                loss, _ = self._back_mod(*front[:6], self.w_dev)
+                self._zero_grads()
+                loss.backward()
+                self._opt_tail()
         torch.cuda.current_stream().wait_stream(s)
 
         self._fill_inputs(self._sample_for_warmup())
         self.gFront = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.gFront):
-            self._gcur, self._h_new, self._ecounts = self._front()
+            (self._nodes, self._uref, self._ei, self._ea, self._seg,
+             self._h_new, self._ecounts) = self._front()
         # capture records without executing — replay once so the static
-        # buffers hold real contents before BACK's capture
+        # buffers hold real contents before the back stage warms/captures
         self.gFront.replay()
-        self.gBack = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.gBack, pool=self.gFront.pool()):
-            self._grads, self._log7 = self._back(self._gcur, self._h_new)
+        # the differentiable stage is captured by torch's supported fwd+bwd
+        # capture (its own pool, its own static grad buffers)
+        self._graphed = torch.cuda.make_graphed_callables(
+            self._back_mod,
+            (self._nodes, self._uref, self._ei, self._ea, self._seg,
+             self._h_new, self.w_dev))
 
     # -------------------------------------------------------------- iter
     def try_iter(self, graph_list):
@@ -339,6 +351,24 @@ class UpdateEngine:
         self.gFront.replay()
         if int(self._ecounts.max().cpu()) > self.E_cap:
             return None
-        self.gBack.replay()
-        self._opt_tail(self._grads)
-        return self._log7.clone()
+        loss, log7 = self._graphed(self._nodes, self._uref, self._ei,
+                                   self._ea, self._seg, self._h_new,
+                                   self.w_dev)
+        self._zero_grads()
+        loss.backward()
+        self._opt_tail()
+        return log7.clone()
+
+
+class _BackCallable(torch.nn.Module):
+    """nn.Module wrapper so make_graphed_callables tracks the CBF and actor
+    parameters; forwards to the engine's differentiable stage."""
+
+    def __init__(self, engine: UpdateEngine):
+        super().__init__()
+        self.cbf = engine.algo.cbf
+        self.actor = engine.algo.actor
+        self._engine = engine
+
+    def forward(self, nodes, uref, ei, ea, seg, h_new, w_dev):
+        return self._engine._back(nodes, uref, ei, ea, seg, h_new, w_dev)

@@ -77,17 +77,25 @@ def eager_loss(gl):
 
 
 gl = e._sample_for_warmup()
+params = [p for p in algo.cbf.parameters() if p.requires_grad] + \
+         [p for p in algo.actor.parameters() if p.requires_grad]
+
+def run_graphed():
+    e.gFront.replay()
+    assert int(e._ecounts.max().cpu()) <= e.E_cap
+    return e._graphed(e._nodes, e._uref, e._ei, e._ea, e._seg, e._h_new,
+                      e.w_dev)
 
 # ---- probe 1: replayed grads vs eager grads on the SAME batch
 e._fill_inputs(gl)
-e.gFront.replay()
-assert int(e._ecounts.max().cpu()) <= e.E_cap
-e.gBack.replay()
-g_cap = [t.clone() for t in e._grads]
-log_cap = e._log7.clone()
+lcap, log_cap = run_graphed()
+e._zero_grads()
+lcap.backward()
+g_cap = [p.grad.clone() for p in params]
+log_cap = log_cap.clone()
 
 loss, parts = eager_loss(gl)
-g_eag = torch.autograd.grad(loss, e._params)
+g_eag = torch.autograd.grad(loss, params)
 print("log7 cap :", [round(float(x), 5) for x in log_cap[:4]])
 print("loss eag :", [round(float(x), 5) for x in parts])
 rows = []
@@ -103,18 +111,19 @@ for cos, n, an, bn in rows[:8]:
 print(f"median cosine: {sorted(r[0] for r in rows)[len(rows)//2]:+.4f}")
 
 # ---- probe 2: does the captured forward see weight changes?
-e._opt_tail(e._grads)
+for p, g in zip(params, g_cap):
+    p.grad = g
+e._opt_tail()
 e._fill_inputs(gl)
-e.gFront.replay()
-e.gBack.replay()
-log_cap2 = e._log7.clone()
+_, log_cap2 = run_graphed()
+log_cap2 = log_cap2.clone()
+with torch.no_grad():
+    pass
 loss2, parts2 = eager_loss(gl)
 print("post-step cap:", [round(float(x), 5) for x in log_cap2[:4]])
 print("post-step eag:", [round(float(x), 5) for x in parts2])
 
 # ---- probe 3: replay determinism (same weights, same inputs)
-e.gFront.replay()
-e.gBack.replay()
-log_cap3 = e._log7.clone()
+_, log_cap3 = run_graphed()
 print("replay determinism maxdiff:",
       (log_cap3 - log_cap2).abs().max().item())
