@@ -329,11 +329,24 @@ class UpdateEngine:
         # buffers hold real contents before the back stage warms/captures
         self.gFront.replay()
         # the differentiable stage is captured by torch's supported fwd+bwd
-        # capture (its own pool, its own static grad buffers)
-        self._graphed = torch.cuda.make_graphed_callables(
-            self._back_mod,
-            (self._nodes, self._uref, self._ei, self._ea, self._seg,
-             self._h_new, self.w_dev))
+        # capture (its own pool, its own static grad buffers).  The custom
+        # fused-linear autograd Function measured corrupted gradients under
+        # graphed capture, so the capture runs the MLPs on plain autocast
+        # GEMMs (launch cost is amortized by the capture anyway); the fused
+        # kernels stay active for the no_grad FRONT graph and eager paths.
+        from .nn.mlp import MLP
+        mlps = [m for mod in (algo.cbf, algo.actor) for m in mod.modules()
+                if isinstance(m, MLP) and getattr(m, "fused_mfma", False)]
+        for m in mlps:
+            m.fused_mfma = False
+        try:
+            self._graphed = torch.cuda.make_graphed_callables(
+                self._back_mod,
+                (self._nodes, self._uref, self._ei, self._ea, self._seg,
+                 self._h_new, self.w_dev))
+        finally:
+            for m in mlps:
+                m.fused_mfma = True
 
     # -------------------------------------------------------------- iter
     def try_iter(self, graph_list):

@@ -18,7 +18,12 @@ def _wrap_forward(module: torch.nn.Module):
 
     @functools.wraps(orig)
     def wrapped(*args, **kwargs):
-        with torch.autocast("cuda", dtype=torch.bfloat16):
+        # cache_enabled=False: the autocast weight-cast cache is documented
+        # as incompatible with CUDA/hipGraph capture (cached casts go stale
+        # across replays); the bf16 weight mirrors already dedup the hot
+        # casts, so the cache buys nothing here
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            cache_enabled=False):
             out = orig(*args, **kwargs)
         return out.float()
 
