@@ -131,9 +131,17 @@ class GCBF(Algorithm):
 
     # -------------------------------------------------------------- training
     def _make_update_engine(self):
-        """Try to build the hipGraph-captured update engine (GPU only)."""
+        """Try to build the hipGraph-captured update engine (GPU only).
+
+        EXPERIMENTAL, opt-in via GCBF_AMD_UPDATE_CAPTURE=1: the captured
+        forward/losses match the eager path, but the captured backward
+        intermittently returns corrupted gradients for a layout-dependent
+        subset of parameters (pool-aliasing in graphed backward under ROCm;
+        see gcbf_amd/update_engine.py and tools/upde_repro.py for the probe
+        trail).  Until that is root-caused the default update path stays
+        eager."""
         self._upd_engine_tried = True
-        if os.environ.get("GCBF_AMD_UPDATE_CAPTURE", "1") == "0":
+        if os.environ.get("GCBF_AMD_UPDATE_CAPTURE", "0") != "1":
             return
         if type(self) is not GCBF:
             return

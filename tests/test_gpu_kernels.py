@@ -643,6 +643,10 @@ def test_rollout_engine_soft_capacity_overflow_fallback():
     assert saw_eager
 
 
+@pytest.mark.skipif(
+    __import__("os").environ.get("GCBF_AMD_UPDATE_CAPTURE") != "1",
+    reason="captured update engine is experimental (opt-in via "
+           "GCBF_AMD_UPDATE_CAPTURE=1)")
 def test_update_engine_matches_eager():
     """The hipGraph-captured update engine must produce the same losses and
     weight updates as the eager inner iterations, from identical state."""
