@@ -177,3 +177,35 @@ def test_hyperparams_table_matches_reference():
                  "loss_safe_coef": 1.0, "loss_h_dot_coef": 0.2}
     assert read_params("SimpleCar", "macbf")["loss_h_dot_coef"] == 1.0
     assert read_params("NoSuchEnv", "gcbf") is None
+
+
+def test_update_engine_not_built_on_cpu():
+    """The captured update engine must gate itself off on CPU and leave the
+    eager path untouched (on_append hook unset, ring ids absent)."""
+    import os
+    import torch
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(0)
+    dev = torch.device("cpu")
+    env = make_env("DubinsCar", 4, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 4, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=20)
+    data = env.reset()
+    os.environ["GCBF_AMD_UPDATE_CAPTURE"] = "1"
+    try:
+        for step in range(1, 21):
+            data.update(u_ref=env.u_ref(data))
+            a = algo.step(data, prob=0.5)
+            data, r, done, info = env.step(a)
+            if done:
+                data = env.reset()
+            if algo.is_update(step):
+                algo.update(step, None)
+    finally:
+        os.environ.pop("GCBF_AMD_UPDATE_CAPTURE", None)
+    assert algo._upd_engine is None
+    assert algo._upd_engine_tried
+    assert algo.buffer.on_append is None
