@@ -3,7 +3,6 @@ checkpoint round-trip, test-time refinement."""
 import os
 
 import numpy as np
-import pytest
 import torch
 
 from gcbf_amd.algo import make_algo

@@ -6,7 +6,6 @@ whose gradients must agree with the average of the per-rank gradients.
 """
 import os
 
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

@@ -2,7 +2,6 @@
 naive per-graph transcription of the reference formulas."""
 import math
 
-import numpy as np
 import pytest
 import torch
 

@@ -3,7 +3,6 @@
 These same tests run against the HIP kernels on GPU (the ops dispatch layer
 routes CUDA tensors to gcbf_amd._C), so they double as kernel numerics tests.
 """
-import math
 
 import pytest
 import torch
