@@ -102,8 +102,13 @@ std::vector<torch::Tensor> segment_attn_bwd(torch::Tensor grad_out,
     CHECK_IN(ptr);
     const int64_t E = msg.size(0), D = msg.size(1);
     const int64_t N = ptr.size(0) - 1;
-    auto dmsg = torch::empty_like(msg);
-    auto dgate = torch::empty({E}, msg.options());
+    // zeros, not empty: the kernel only writes edges covered by a CSR
+    // segment, and padded edge buffers (sentinel segment id past the last
+    // node) legitimately contain uncovered edges — leaving those rows
+    // uninitialized poisoned every gradient downstream of the aggregation
+    // in the captured update engine
+    auto dmsg = torch::zeros_like(msg);
+    auto dgate = torch::zeros({E}, msg.options());
     if (N > 0)
         launch_seg_attn_bwd(grad_out.data_ptr<float>(), msg.data_ptr<float>(),
                             att.data_ptr<float>(), ptr.data_ptr<int>(),

