@@ -304,6 +304,30 @@ class UpdateEngine:
         algo = self.algo
         self._back_mod = _BackCallable(self)
 
+        # the whole engine (both captures + warmup) runs the MLPs on plain
+        # autocast GEMMs: (a) the custom fused-linear Function measured
+        # corrupted gradients under graphed capture, and (b) FRONT and BACK
+        # must be numerically IDENTICAL per row — the ḣ residue divides an
+        # h_new−h_next difference by dt (~33x amplification), so a
+        # fused-vs-hipBLASLt mismatch between the two stages shifts the
+        # re-link residue.  Launch cost is amortized by the capture anyway;
+        # the fused kernels stay active for rollout and eager paths.
+        from .nn.mlp import MLP
+        self._mlps = [m for mod in (algo.cbf, algo.actor)
+                      for m in mod.modules()
+                      if isinstance(m, MLP) and getattr(m, "fused_mfma",
+                                                        False)]
+        for m in self._mlps:
+            m.fused_mfma = False
+        try:
+            self._build_captures()
+        finally:
+            # restore the fused path for rollout / eager consumers (the
+            # captured graphs recorded the plain-GEMM kernels already)
+            for m in self._mlps:
+                m.fused_mfma = True
+
+    def _build_captures(self):
         # warmup: real iterations over the engine's padded shapes (warms
         # hipBLASLt shape caches, materializes Adam state + bf16 mirrors)
         s = torch.cuda.Stream()
@@ -329,24 +353,11 @@ class UpdateEngine:
         # buffers hold real contents before the back stage warms/captures
         self.gFront.replay()
         # the differentiable stage is captured by torch's supported fwd+bwd
-        # capture (its own pool, its own static grad buffers).  The custom
-        # fused-linear autograd Function measured corrupted gradients under
-        # graphed capture, so the capture runs the MLPs on plain autocast
-        # GEMMs (launch cost is amortized by the capture anyway); the fused
-        # kernels stay active for the no_grad FRONT graph and eager paths.
-        from .nn.mlp import MLP
-        mlps = [m for mod in (algo.cbf, algo.actor) for m in mod.modules()
-                if isinstance(m, MLP) and getattr(m, "fused_mfma", False)]
-        for m in mlps:
-            m.fused_mfma = False
-        try:
-            self._graphed = torch.cuda.make_graphed_callables(
-                self._back_mod,
-                (self._nodes, self._uref, self._ei, self._ea, self._seg,
-                 self._h_new, self.w_dev))
-        finally:
-            for m in mlps:
-                m.fused_mfma = True
+        # capture (its own pool, its own static grad buffers)
+        self._graphed = torch.cuda.make_graphed_callables(
+            self._back_mod,
+            (self._nodes, self._uref, self._ei, self._ea, self._seg,
+             self._h_new, self.w_dev))
 
     # -------------------------------------------------------------- iter
     def try_iter(self, graph_list):
