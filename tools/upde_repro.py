@@ -6,6 +6,7 @@ import os
 
 import torch
 
+os.environ["GCBF_AMD_UPDATE_CAPTURE"] = "1"      # engine is opt-in
 os.environ["GCBF_AMD_UPDATE_CAPTURE_DEBUG"] = "1"
 
 from gcbf_amd.env import make_env
