@@ -83,7 +83,8 @@ class CBFGNNLayer(nn.Module):
         if node_mask is not None:
             # int = "first n rows" static slice (capture-safe); tensor =
             # boolean mask (batched graphs)
-            gamma_in = gamma_in[:node_mask] if isinstance(node_mask, int)                 else gamma_in[node_mask]
+            gamma_in = (gamma_in[:node_mask] if isinstance(node_mask, int)
+                        else gamma_in[node_mask])
         return self.gamma(gamma_in)
 
     def attention(self, data: GraphBatch) -> Tensor:
@@ -119,7 +120,8 @@ class ControllerGNNLayer(nn.Module):
         aggr = self.aggr_module(msg, dst, num_nodes)
         gamma_in = torch.cat([aggr, x], dim=1)
         if node_mask is not None:
-            gamma_in = gamma_in[:node_mask] if isinstance(node_mask, int)                 else gamma_in[node_mask]
+            gamma_in = (gamma_in[:node_mask] if isinstance(node_mask, int)
+                        else gamma_in[node_mask])
         return self.gamma(gamma_in)
 
 
