@@ -58,7 +58,9 @@ class CBFGNN(nn.Module):
 
     def forward(self, data: GraphBatch) -> Tensor:
         nm = data.agent_mask
-        if data.agents_first_n is not None:
+        if data.agent_index is not None:
+            nm = data.agent_index      # static LONG indices (capture-safe)
+        elif data.agents_first_n is not None:
             nm = data.agents_first_n
         x = self.feat_transformer.module_0(
             data.x, data.edge_attr, data.edge_index,

@@ -44,7 +44,7 @@ class GraphBatch:
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
                  "u_ref", "_ptr", "_dst_ptr", "seg_dst", "agents_first_n",
-                 "ring_id")
+                 "ring_id", "agent_index")
 
     def __init__(
             self,
@@ -79,6 +79,10 @@ class GraphBatch:
         # optional: slot id in the update engine's device ring (stamped by
         # Buffer.on_append when the captured update engine is active)
         self.ring_id = None
+        # optional: LONG index tensor of agent rows (uniform batches have a
+        # static agent layout) — models prefer it over the boolean
+        # agent_mask because integer indexing is hipGraph-capturable
+        self.agent_index = None
 
     # ------------------------------------------------------------------ sizes
     @property

@@ -139,3 +139,45 @@ def test_mlp_orthogonal_init():
     w = m.net[0].weight  # (64, 32): columns orthonormal
     assert torch.allclose(w.t() @ w, torch.eye(32), atol=1e-5)
     assert (m.net[0].bias == 0).all()
+
+
+def test_agent_index_matches_boolean_mask():
+    """LONG agent_index indexing (hipGraph-capturable) must equal the
+    boolean agent_mask path on a batched graph with obstacle rows."""
+    import torch
+    from gcbf_amd.algo.gcbf import CBFGNN
+    from gcbf_amd.controller import GNNController
+    from gcbf_amd.graph import GraphBatch
+
+    torch.manual_seed(3)
+    B, n, n_obs = 3, 5, 4
+    N = n + n_obs
+    x = torch.cat([torch.zeros(n, 4), torch.ones(n_obs, 4)]).repeat(B, 1)
+    states = torch.randn(B * N, 4)
+    # dense-ish random edges into agent receivers, dst-sorted per graph
+    ei = []
+    for b in range(B):
+        for i in range(n):
+            for j in range(N):
+                if j != i and torch.rand(1).item() < 0.6:
+                    ei.append((b * N + j, b * N + i))
+    ei = torch.tensor(ei, dtype=torch.long).t()
+    ea = torch.randn(ei.shape[1], 5)
+    am = torch.zeros(B * N, dtype=torch.bool)
+    am[(torch.arange(B * N) % N) < n] = True
+    ptr = torch.arange(B + 1, dtype=torch.long) * N
+
+    g_bool = GraphBatch(x=x, pos=states[:, :2], states=states,
+                        edge_index=ei, edge_attr=ea, agent_mask=am,
+                        u_ref=torch.randn(B * n, 2), ptr=ptr)
+    g_idx = GraphBatch(x=x, pos=states[:, :2], states=states,
+                       edge_index=ei, edge_attr=ea, agent_mask=am,
+                       u_ref=g_bool.u_ref, ptr=ptr)
+    g_idx.agent_index = am.nonzero().flatten()
+
+    cbf = CBFGNN(num_agents=n, node_dim=4, edge_dim=5, phi_dim=256).eval()
+    actor = GNNController(num_agents=n, node_dim=4, edge_dim=5, phi_dim=256,
+                          action_dim=2).eval()
+    with torch.no_grad():
+        assert torch.equal(cbf(g_bool), cbf(g_idx))
+        assert torch.equal(actor(g_bool), actor(g_idx))
