@@ -151,11 +151,10 @@ class GCBF(Algorithm):
             return
         env = self._env
         data = env.data
-        # v1 requires every node to be an agent (no obstacle points): an
-        # all-agent graph makes the boolean agent mask a mathematical no-op,
-        # so the captured pipeline can drop it (boolean indexing is not
-        # capturable)
-        if (data is None or data.num_nodes != env.num_agents
+        # obstacle nodes are supported via the static agent-index layout
+        # (agents are the first n rows of every graph); the MACBF
+        # neighbor-cap path stays eager
+        if (data is None
                 or getattr(env, "_max_neighbors", None) is not None):
             return
         try:

@@ -95,6 +95,10 @@ class DubinsCar(SimpleCar):
         ctrl = torch.zeros(n_nodes, 2, dtype=s.dtype, device=s.device)
         if agent_mask is None:
             ctrl = torch.stack([u[:, 0] * 10, u[:, 1]], dim=1)
+        elif data.agent_index is not None:  # capture-safe integer scatter
+            ctrl = ctrl.index_copy(
+                0, data.agent_index,
+                torch.stack([u[:, 0] * 10, u[:, 1]], dim=1))
         else:
             ctrl = ctrl.masked_scatter(
                 agent_mask.unsqueeze(1).expand(-1, 2),
@@ -276,7 +280,12 @@ class DubinsCar(SimpleCar):
     def u_ref(self, data: GraphBatch) -> Tensor:
         # PID heading/accel controller (reference gcbf/env/dubins_car.py:764-816)
         am = data.agent_mask
-        states = data.states[am] if am is not None else data.states
+        if data.agent_index is not None:   # capture-safe integer indexing
+            states = data.states.index_select(0, data.agent_index)
+        elif am is not None:
+            states = data.states[am]
+        else:
+            states = data.states
         states = states.reshape(-1, self.num_agents, self.state_dim)
         diff = (states - self._goal).reshape(-1, self.state_dim)
         states = states.reshape(-1, self.state_dim)

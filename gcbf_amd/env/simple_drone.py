@@ -106,8 +106,11 @@ class SimpleDrone(MultiAgentEnv):
         mask_f = am.to(s.dtype).unsqueeze(1)
         xdot = xdot * mask_f  # obstacles static
         ctrl = torch.zeros_like(xdot)
-        ctrl = ctrl.masked_scatter(
-            am.unsqueeze(1).expand(-1, self.state_dim), u @ self._B.t())
+        if data.agent_index is not None:    # capture-safe integer scatter
+            ctrl = ctrl.index_copy(0, data.agent_index, u @ self._B.t())
+        else:
+            ctrl = ctrl.masked_scatter(
+                am.unsqueeze(1).expand(-1, self.state_dim), u @ self._B.t())
         xdot = xdot + ctrl
         if s.shape[0] == self.num_agents + self._obs.shape[0]:
             agent_states = s[am]
@@ -262,7 +265,12 @@ class SimpleDrone(MultiAgentEnv):
     def u_ref(self, data: GraphBatch) -> Tensor:
         # reference gcbf/env/simple_drone.py:349-377
         am = data.agent_mask
-        states = data.states[am] if am is not None else data.states
+        if data.agent_index is not None:   # capture-safe integer indexing
+            states = data.states.index_select(0, data.agent_index)
+        elif am is not None:
+            states = data.states[am]
+        else:
+            states = data.states
         states = states.reshape(-1, self.num_agents, self.state_dim)
         diff = states - self._goal
 

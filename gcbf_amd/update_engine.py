@@ -73,8 +73,9 @@ class UpdateEngine:
         self.device = env.device
 
         data = env.data
-        assert data is not None and data.num_nodes == env.num_agents
-        self.N = data.num_nodes            # nodes per graph (= agents here)
+        assert data is not None
+        self.N = data.num_nodes            # nodes per graph (incl. obstacles)
+        self.n = env.num_agents            # agent rows come first per graph
         self.S = env.state_dim
         self.pd = 3 if env.state_dim == 6 else 2
         self.nd = env.node_dim
@@ -84,6 +85,7 @@ class UpdateEngine:
         seg_len = 3
         self.G_cap = seg_len * (algo.batch_size // 5)
         self.Ntot = self.G_cap * self.N
+        self.nA = self.G_cap * self.n      # agent rows in the padded batch
         if algo.buffer.size == 0:
             raise RuntimeError("empty buffer at engine init")
 
@@ -93,7 +95,7 @@ class UpdateEngine:
         # wasted FLOPs — stay modest (4x padding measured 3x slower updates)
         mean_e = max(1.0, sum(g.num_edges for g in algo.buffer.data)
                      / algo.buffer.size)
-        full = self.G_cap * self.N * (self.N - 1)
+        full = self.G_cap * self.n * (self.N - 1)   # agents receive only
         want = int(1.35 * mean_e * self.G_cap) + 1024
         self.E_cap = min(full, (want + 2047) // 2048 * 2048)
 
@@ -101,7 +103,7 @@ class UpdateEngine:
         # ---- device ring of (states, u_ref) per appended graph
         self.CAP = algo.buffer.MAX_SIZE + 2 * algo.batch_size
         self.ring_states = torch.zeros(self.CAP, self.N, self.S, device=dev)
-        self.ring_uref = torch.zeros(self.CAP, self.N, self.ad, device=dev)
+        self.ring_uref = torch.zeros(self.CAP, self.n, self.ad, device=dev)
         self.next_id = 0
 
         # ---- static tiles / batch skeleton
@@ -111,6 +113,18 @@ class UpdateEngine:
                                 device=dev) * self.N
         self.ptr2 = torch.arange(2 * self.G_cap + 1, dtype=torch.long,
                                  device=dev) * self.N
+        # static agent layout: rows [g*N, g*N + n) are the agents of graph g
+        base = torch.arange(self.G_cap, device=dev).unsqueeze(1) * self.N
+        self.agent_index = (base + torch.arange(self.n, device=dev)
+                            ).reshape(-1)
+        self.agent_index2 = torch.cat([self.agent_index,
+                                       self.agent_index + self.Ntot])
+        if self.N == self.n:
+            self.agent_mask_tile = None
+        else:
+            am = torch.zeros(self.N, dtype=torch.bool, device=dev)
+            am[: self.n] = True
+            self.agent_mask_tile = am.repeat(self.G_cap)
 
         # ---- per-iteration inputs (content updated before each replay)
         self.idx_host = torch.empty(self.G_cap, dtype=torch.long,
@@ -146,19 +160,21 @@ class UpdateEngine:
         nodes = self.ring_states.index_select(
             0, self.idx_dev).reshape(self.Ntot, self.S)
         uref = self.ring_uref.index_select(
-            0, self.idx_dev).reshape(self.Ntot, self.ad)
+            0, self.idx_dev).reshape(self.nA, self.ad)
         return nodes, uref
 
     def _build_cur(self, nodes, uref):
         env = self.env
         ei, seg, ea, ecount = self._ext.build_graph_padded(
-            nodes[:, :self.pd].contiguous(), nodes, self.G_cap, self.N,
+            nodes[:, :self.pd].contiguous(), nodes, self.G_cap, self.n,
             env.params["comm_radius"], -1, env._attr_kind, self.ed,
             self.E_cap)
         gcur = GraphBatch(x=self.x_tile, pos=nodes[:, :self.pd],
                           states=nodes, edge_index=ei, edge_attr=ea,
+                          agent_mask=self.agent_mask_tile,
                           u_ref=uref, ptr=self.ptr)
         gcur.seg_dst = seg
+        gcur.agent_index = self.agent_index
         return gcur, ecount
 
     def _front(self):
@@ -173,11 +189,13 @@ class UpdateEngine:
             ns = gnext.states
             ei2, seg2, ea2, ecount2 = self._ext.build_graph_padded(
                 ns[:, :self.pd].contiguous(), ns.contiguous(), self.G_cap,
-                self.N, env.params["comm_radius"], -1, env._attr_kind,
+                self.n, env.params["comm_radius"], -1, env._attr_kind,
                 self.ed, self.E_cap)
             grel = GraphBatch(x=self.x_tile, pos=ns[:, :self.pd], states=ns,
-                              edge_index=ei2, edge_attr=ea2, ptr=self.ptr)
+                              edge_index=ei2, edge_attr=ea2,
+                              agent_mask=self.agent_mask_tile, ptr=self.ptr)
             grel.seg_dst = seg2
+            grel.agent_index = self.agent_index
             h_new = algo.cbf(grel)
             ecounts = torch.stack([ecount[0], ecount2[0]])
         return (gcur.states, gcur.u_ref, gcur.edge_index, gcur.edge_attr,
@@ -194,10 +212,13 @@ class UpdateEngine:
 
         gcur = GraphBatch(x=self.x_tile, pos=nodes[:, :self.pd],
                           states=nodes, edge_index=ei, edge_attr=ea,
+                          agent_mask=self.agent_mask_tile,
                           u_ref=uref, ptr=self.ptr)
         gcur.seg_dst = seg
+        gcur.agent_index = self.agent_index
         actions = algo.actor(gcur)
         gnext = env.forward_graph(gcur, actions)
+        gnext.agent_index = self.agent_index
 
         states2 = torch.cat([gcur.states, gnext.states], dim=0)
         ea2 = torch.cat([gcur.edge_attr, gnext.edge_attr], dim=0)
@@ -210,14 +231,18 @@ class UpdateEngine:
                           seg + self.Ntot])
         both = GraphBatch(x=self.x2_tile, pos=states2[:, :self.pd],
                           states=states2, edge_index=ei2, edge_attr=ea2,
+                          agent_mask=None if self.agent_mask_tile is None
+                          else self.agent_mask_tile.repeat(2),
                           ptr=self.ptr2)
         both.seg_dst = seg2
+        both.agent_index = self.agent_index2
         h_both = algo.cbf(both)
-        h, h_next = h_both[:self.Ntot], h_both[self.Ntot:]
+        h, h_next = h_both[:self.nA], h_both[self.nA:]
         hv = h[:, 0]
 
+        # per-AGENT-row weights (h/actions/masks are per agent)
         w_node = w_dev.view(self.G_cap, 1).expand(
-            self.G_cap, self.N).reshape(self.Ntot)
+            self.G_cap, self.n).reshape(self.nA)
         cw = w_node.sum()
 
         # identical weighted-sum forms as GCBF._iter_eager, with the pad

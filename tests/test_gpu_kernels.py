@@ -762,3 +762,36 @@ def test_update_engine_matches_eager():
     # and a full engine iteration must run end to end with finite logs
     out = e.try_iter(e._sample_for_warmup())
     assert out is not None and torch.isfinite(out).all()
+
+
+@pytest.mark.skipif(
+    __import__("os").environ.get("GCBF_AMD_UPDATE_CAPTURE") != "1",
+    reason="captured update engine is experimental (opt-in via "
+           "GCBF_AMD_UPDATE_CAPTURE=1)")
+def test_update_engine_with_obstacles():
+    """Captured update engine on a scene WITH obstacle nodes (static
+    agent-index layout): engine must build and run finite iterations."""
+    from gcbf_amd.env import make_env
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.rollout import RolloutEngine
+    from gcbf_amd.trainer.utils import set_seed
+    from gcbf_amd.utils.amp import enable_bf16
+
+    set_seed(3)
+    dev = torch.device("cuda")
+    e0 = make_env("DubinsCar", 16, dev)
+    p = e0.default_params
+    p["num_obs"] = 4
+    env = make_env("DubinsCar", 16, dev, params=p)
+    env.train()
+    algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=256)
+    enable_bf16(algo)
+    env.reset()
+    eng = RolloutEngine(env, algo)
+    for _ in range(256):
+        if eng.step(prob=0.7):
+            eng.reload()
+    out = algo.update(256)
+    assert algo._upd_engine is not None, "engine must build with obstacles"
+    assert all(0 <= v <= 1 for v in out.values())
