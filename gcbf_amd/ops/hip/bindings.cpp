@@ -82,7 +82,10 @@ std::vector<torch::Tensor> segment_attn_fwd(torch::Tensor msg,
     TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
     const int64_t E = msg.size(0), D = msg.size(1);
     const int64_t N = ptr.size(0) - 1;
-    auto att = torch::empty({E}, msg.options());
+    // att is zero-initialized for the same reason as the backward outputs:
+    // edges past the last CSR segment (sentinel pads) are never written by
+    // the kernel and must not surface uninitialized memory
+    auto att = torch::zeros({E}, msg.options());
     auto out = torch::empty({N, D}, msg.options());
     if (N > 0)
         launch_seg_attn_fwd(msg.data_ptr<float>(), gate.data_ptr<float>(),
