@@ -107,6 +107,10 @@ class GCBF(Algorithm):
         # DP hook: called after backward, before the optimizer steps
         self.grad_sync: Optional[Callable[[], None]] = None
 
+        # device ring of replayed states (fast batched re-batching for the
+        # eager update path; also backs the captured update engine)
+        self._ring = None
+        self._ring_tried = False
         # hipGraph-captured update engine (created lazily at first update
         # when supported); None -> eager inner iterations
         self._upd_engine = None
@@ -132,6 +136,29 @@ class GCBF(Algorithm):
         return step % self.batch_size == 0
 
     # -------------------------------------------------------------- training
+    def _make_ring(self):
+        """Device ring + batched exact rebuild for sampled batches (replaces
+        per-graph Python concatenation in the update; see gcbf_amd/ring.py).
+        """
+        self._ring_tried = True
+        if os.environ.get("GCBF_AMD_RING", "1") == "0":
+            return
+        env = self._env
+        if (env.data is None
+                or getattr(env, "_max_neighbors", None) is not None):
+            return
+        try:
+            from ..ring import RingStore
+            ring = RingStore(env, Buffer.MAX_SIZE + 2 * self.batch_size)
+            for g in list(self.buffer.data) + list(self.memory.data):
+                ring.push(g)
+            self.buffer.on_append = ring.push
+            self._ring = ring
+        except Exception as e:
+            import warnings
+            warnings.warn(f"ring batcher unavailable ({e})")
+            self._ring = None
+
     def _make_update_engine(self):
         """Try to build the hipGraph-captured update engine (GPU only).
 
@@ -186,6 +213,8 @@ class GCBF(Algorithm):
         else:
             _tick = None
 
+        if self._ring is None and not self._ring_tried:
+            self._make_ring()
         if self._upd_engine is None and not self._upd_engine_tried:
             self._make_update_engine()
 
@@ -226,7 +255,10 @@ class GCBF(Algorithm):
         alpha = self.params["alpha"]
         t0 = _tick() if prof else 0
 
-        graphs = GraphBatch.from_list(graph_list)
+        if self._ring is not None and self._ring.usable(graph_list):
+            graphs = self._ring.batch(graph_list)
+        else:
+            graphs = GraphBatch.from_list(graph_list)
         graphs.edge_attr.requires_grad_(True)
         if prof:
             t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1

@@ -100,11 +100,12 @@ class UpdateEngine:
         self.E_cap = min(full, (want + 2047) // 2048 * 2048)
 
         dev = self.device
-        # ---- device ring of (states, u_ref) per appended graph
-        self.CAP = algo.buffer.MAX_SIZE + 2 * algo.batch_size
-        self.ring_states = torch.zeros(self.CAP, self.N, self.S, device=dev)
-        self.ring_uref = torch.zeros(self.CAP, self.n, self.ad, device=dev)
-        self.next_id = 0
+        # ---- shared device ring (owned by the algo; also feeds the fast
+        # eager batcher, gcbf_amd/ring.py)
+        self.ring = algo._ring
+        if self.ring is None:
+            raise RuntimeError("ring store unavailable")
+        assert self.ring.N == self.N and self.ring.n == self.n
 
         # ---- static tiles / batch skeleton
         self.x_tile = data.x.repeat(self.G_cap, 1).contiguous()
@@ -138,28 +139,13 @@ class UpdateEngine:
         self.goal_static = env._goal.clone().contiguous()
         env._goal = self.goal_static
 
-        # backfill graphs appended before the engine existed, then hook
-        for g in list(algo.buffer.data) + list(algo.memory.data):
-            self.push(g)
-        algo.buffer.on_append = self.push
-
         self._build()
-
-    # ------------------------------------------------------------- ring
-    def push(self, g: GraphBatch):
-        if g.ring_id is not None:
-            return
-        slot = self.next_id % self.CAP
-        self.ring_states[slot].copy_(g.states, non_blocking=True)
-        self.ring_uref[slot].copy_(g.u_ref, non_blocking=True)
-        g.ring_id = self.next_id
-        self.next_id += 1
 
     # ------------------------------------------------------------- bodies
     def _gather(self):
-        nodes = self.ring_states.index_select(
+        nodes = self.ring.states.index_select(
             0, self.idx_dev).reshape(self.Ntot, self.S)
-        uref = self.ring_uref.index_select(
+        uref = self.ring.uref.index_select(
             0, self.idx_dev).reshape(self.nA, self.ad)
         return nodes, uref
 
@@ -288,10 +274,11 @@ class UpdateEngine:
     # ------------------------------------------------------------ capture
     def _fill_inputs(self, graph_list):
         L = len(graph_list)
+        cap = self.ring.CAP
         ids = [g.ring_id for g in graph_list]
-        self.idx_host[:L] = torch.tensor([i % self.CAP for i in ids],
+        self.idx_host[:L] = torch.tensor([i % cap for i in ids],
                                          dtype=torch.long)
-        self.idx_host[L:] = ids[0] % self.CAP
+        self.idx_host[L:] = ids[0] % cap
         self.w_host[:L] = 1.0
         self.w_host[L:] = 0.0
         self.idx_dev.copy_(self.idx_host, non_blocking=True)

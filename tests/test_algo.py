@@ -207,4 +207,50 @@ def test_update_engine_not_built_on_cpu():
         os.environ.pop("GCBF_AMD_UPDATE_CAPTURE", None)
     assert algo._upd_engine is None
     assert algo._upd_engine_tried
-    assert algo.buffer.on_append is None
+    # the ring batcher MAY hook on_append (it works on CPU); the captured
+    # engine must not have been built
+    assert (algo.buffer.on_append is None
+            or algo.buffer.on_append == getattr(algo._ring, "push", None))
+
+
+def test_ring_batch_matches_from_list():
+    """RingStore.batch must reproduce GraphBatch.from_list exactly on
+    sampled training graphs (states, u_ref, edges, attrs)."""
+    import torch
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.graph import GraphBatch
+    from gcbf_amd.trainer.utils import set_seed
+
+    set_seed(2)
+    dev = torch.device("cpu")
+    env = make_env("DubinsCar", 8, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 8, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=40)
+    data = env.reset()
+    for step in range(20):   # pre-ring appends (exercise backfill)
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.5)
+        data, r, done, info = env.step(a)
+        data = env.reset() if done else data
+    algo._make_ring()
+    assert algo._ring is not None
+    for step in range(20):   # post-ring appends (exercise the hook)
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.5)
+        data, r, done, info = env.step(a)
+        data = env.reset() if done else data
+
+    gl = algo.buffer.sample(8, 3)
+    assert algo._ring.usable(gl)
+    fast = algo._ring.batch(gl)
+    ref = GraphBatch.from_list(gl)
+    assert torch.equal(fast.states, ref.states)
+    assert torch.equal(fast.u_ref, ref.u_ref)
+    assert torch.equal(fast.edge_index, ref.edge_index)
+    assert torch.allclose(fast.edge_attr, ref.edge_attr, atol=1e-6)
+    assert torch.equal(fast.x, ref.x)
+    assert fast.num_graphs == ref.num_graphs
+    if ref.agent_mask is not None:
+        assert torch.equal(fast.agent_mask, ref.agent_mask)
