@@ -255,9 +255,16 @@ class GCBF(Algorithm):
         alpha = self.params["alpha"]
         t0 = _tick() if prof else 0
 
+        graphs = None
         if self._ring is not None and self._ring.usable(graph_list):
-            graphs = self._ring.batch(graph_list)
-        else:
+            try:
+                graphs = self._ring.batch(graph_list)
+            except Exception as e:
+                import warnings
+                warnings.warn(f"ring batch failed ({e}); "
+                              f"falling back to from_list")
+                self._ring = None
+        if graphs is None:
             graphs = GraphBatch.from_list(graph_list)
         graphs.edge_attr.requires_grad_(True)
         if prof:

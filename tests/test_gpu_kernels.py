@@ -795,3 +795,36 @@ def test_update_engine_with_obstacles():
     out = algo.update(256)
     assert algo._upd_engine is not None, "engine must build with obstacles"
     assert all(0 <= v <= 1 for v in out.values())
+
+
+def test_ring_batch_matches_from_list_gpu():
+    """RingStore.batch (HIP batched rebuild) must equal from_list on GPU:
+    the stored edges came from the same deterministic kernels."""
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.graph import GraphBatch
+    from gcbf_amd.rollout import RolloutEngine
+    from gcbf_amd.trainer.utils import set_seed
+
+    set_seed(7)
+    dev = torch.device("cuda")
+    env = make_env("DubinsCar", 16, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=64)
+    env.reset()
+    eng = RolloutEngine(env, algo)
+    for _ in range(64):
+        if eng.step(prob=0.6):
+            eng.reload()
+    algo._make_ring()
+    assert algo._ring is not None
+    gl = algo.buffer.sample(12, 3)
+    assert algo._ring.usable(gl)
+    fast = algo._ring.batch(gl)
+    ref = GraphBatch.from_list(gl)
+    assert torch.equal(fast.states, ref.states)
+    assert torch.equal(fast.edge_index, ref.edge_index)
+    assert torch.allclose(fast.edge_attr, ref.edge_attr, atol=1e-6), \
+        (fast.edge_attr - ref.edge_attr).abs().max()
+    assert torch.equal(fast.u_ref, ref.u_ref)
