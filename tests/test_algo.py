@@ -213,9 +213,14 @@ def test_update_engine_not_built_on_cpu():
             or algo.buffer.on_append == getattr(algo._ring, "push", None))
 
 
-def test_ring_batch_matches_from_list():
+import pytest
+
+
+@pytest.mark.parametrize("obs", [0, 4])
+def test_ring_batch_matches_from_list(obs):
     """RingStore.batch must reproduce GraphBatch.from_list exactly on
-    sampled training graphs (states, u_ref, edges, attrs)."""
+    sampled training graphs (states, u_ref, edges, attrs), with and
+    without obstacle nodes."""
     import torch
     from gcbf_amd.algo import make_algo
     from gcbf_amd.env import make_env
@@ -224,7 +229,10 @@ def test_ring_batch_matches_from_list():
 
     set_seed(2)
     dev = torch.device("cpu")
-    env = make_env("DubinsCar", 8, dev)
+    e0 = make_env("DubinsCar", 8, dev)
+    p = e0.default_params
+    p["num_obs"] = obs
+    env = make_env("DubinsCar", 8, dev, params=p)
     env.train()
     algo = make_algo("gcbf", env, 8, env.node_dim, env.edge_dim,
                      env.action_dim, dev, batch_size=40)
