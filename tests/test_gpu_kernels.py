@@ -797,7 +797,8 @@ def test_update_engine_with_obstacles():
     assert all(0 <= v <= 1 for v in out.values())
 
 
-def test_ring_batch_matches_from_list_gpu():
+@pytest.mark.parametrize("obs", [0, 6])
+def test_ring_batch_matches_from_list_gpu(obs):
     """RingStore.batch (HIP batched rebuild) must equal from_list on GPU:
     the stored edges came from the same deterministic kernels."""
     from gcbf_amd.algo import make_algo
@@ -808,7 +809,10 @@ def test_ring_batch_matches_from_list_gpu():
 
     set_seed(7)
     dev = torch.device("cuda")
-    env = make_env("DubinsCar", 16, dev)
+    e0 = make_env("DubinsCar", 16, dev)
+    p = e0.default_params
+    p["num_obs"] = obs
+    env = make_env("DubinsCar", 16, dev, params=p)
     env.train()
     algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
                      env.action_dim, dev, batch_size=64)
