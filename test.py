@@ -51,7 +51,10 @@ def test(args):
     try:
         settings = read_settings(args.path)
     except TypeError:
-        settings = {"algo": "nominal", "num_agents": args.num_agents}
+        # nominal-controller runs need no checkpoint dir (reference
+        # test.py:29-30); fall back to CLI values with sane defaults
+        settings = {"algo": "nominal",
+                    "num_agents": args.num_agents or 16}
 
     env_name = settings.get("env") if args.env is None else args.env
     num_agents = settings["num_agents"] if args.num_agents is None \
