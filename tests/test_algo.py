@@ -216,11 +216,14 @@ def test_update_engine_not_built_on_cpu():
 import pytest
 
 
-@pytest.mark.parametrize("obs", [0, 4])
-def test_ring_batch_matches_from_list(obs):
+@pytest.mark.parametrize("env_name,obs", [("DubinsCar", 0),
+                                          ("DubinsCar", 4),
+                                          ("SimpleCar", None),
+                                          ("SimpleDrone", None)])
+def test_ring_batch_matches_from_list(env_name, obs):
     """RingStore.batch must reproduce GraphBatch.from_list exactly on
-    sampled training graphs (states, u_ref, edges, attrs), with and
-    without obstacle nodes."""
+    sampled training graphs (states, u_ref, edges, attrs), across env
+    families and with/without obstacle nodes."""
     import torch
     from gcbf_amd.algo import make_algo
     from gcbf_amd.env import make_env
@@ -229,10 +232,11 @@ def test_ring_batch_matches_from_list(obs):
 
     set_seed(2)
     dev = torch.device("cpu")
-    e0 = make_env("DubinsCar", 8, dev)
+    e0 = make_env(env_name, 8, dev)
     p = e0.default_params
-    p["num_obs"] = obs
-    env = make_env("DubinsCar", 8, dev, params=p)
+    if obs is not None:
+        p["num_obs"] = obs
+    env = make_env(env_name, 8, dev, params=p)
     env.train()
     algo = make_algo("gcbf", env, 8, env.node_dim, env.edge_dim,
                      env.action_dim, dev, batch_size=40)
