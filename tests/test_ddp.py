@@ -13,7 +13,15 @@ import torch.multiprocessing as mp
 from gcbf_amd.parallel import GradSynchronizer, broadcast_modules
 
 
-BASE_PORT = 29000 + (os.getpid() % 900)
+def _free_port() -> int:
+    """OS-assigned free TCP port (PID-derived ports collided across
+    sequential pytest runs via TIME_WAIT reuse)."""
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
 
 
 def _init(rank, world, port):
@@ -42,7 +50,8 @@ def _worker_grad_sync(rank, world, q, port):
 def test_grad_synchronizer_averages():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_worker_grad_sync, args=(r, 2, q, BASE_PORT))
+    port = _free_port()
+    ps = [ctx.Process(target=_worker_grad_sync, args=(r, 2, q, port))
           for r in range(2)]
     for p in ps:
         p.start()
@@ -90,8 +99,9 @@ def _worker_gcbf_dp(rank, world, q, port):
 def test_gcbf_dp_two_ranks_stay_in_sync():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
+    port = _free_port()
     ps = [ctx.Process(target=_worker_gcbf_dp,
-                      args=(r, 2, q, BASE_PORT + 7))
+                      args=(r, 2, q, port))
           for r in range(2)]
     for p in ps:
         p.start()
