@@ -100,7 +100,15 @@ class MultiAgentEnv(ABC):
     def reset(self) -> GraphBatch: ...
 
     @abstractmethod
-    def step(self, action: Tensor) -> Tuple[GraphBatch, np.ndarray, bool, dict]: ...
+    def step(self, action: Tensor) -> Tuple[GraphBatch, np.ndarray, bool, dict]:
+        """One env step: (next_data, reward, done, info).
+
+        Info-dict contract (differs from the reference): ``info["collision"]``
+        and ``info["reach"]`` are per-agent BOOLEAN MASK tensors on device,
+        not lists of agent indices (the reference DubinsCar returned
+        indices, gcbf/env/dubins_car.py:612-615).  Boolean masks keep the
+        step host-sync-free; index consumers use ``mask.nonzero()``.
+        """
 
     @abstractmethod
     def forward_graph(self, data: GraphBatch, action: Tensor) -> GraphBatch:

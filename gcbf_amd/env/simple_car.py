@@ -157,6 +157,10 @@ class SimpleCar(MultiAgentEnv):
             agent_mask=self._data.agent_mask)
         if data.agent_mask is not None:
             data.agents_first_n = self.num_agents
+            # keep the obstacle-state view fresh (the eager step path
+            # refreshes it; only its shape is consumed today, but stale
+            # values are a trap for future readers)
+            self._obs = new_states[self.num_agents:]
         self._data = self.add_communication_links(data)
         self._data.u_ref = u_ref_next
         done = bool(self._t >= self.max_episode_steps or reach.all())

@@ -155,9 +155,15 @@ class GraphBatch:
 
         Equivalent of ``Batch.from_data_list`` (gcbf/algo/gcbf.py:159).
         """
+        assert len(graphs) > 0, "from_list needs at least one graph"
         if len(graphs) == 1:
+            # clone edge_attr: callers (GCBF._iter_eager) set
+            # requires_grad_ on the batch's edge_attr, which must not leak
+            # back into the stored replay-buffer graph
             g = graphs[0]
-            return g.replace()
+            out = g.replace(edge_attr=None if g.edge_attr is None
+                            else g.edge_attr.clone())
+            return out
         device = graphs[0].device
         node_counts = [g.num_nodes for g in graphs]
         offsets = torch.zeros(len(graphs) + 1, dtype=torch.long,

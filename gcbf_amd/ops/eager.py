@@ -116,7 +116,10 @@ def dense_radius_graph(
     if max_neighbors is not None and max_neighbors < dist.shape[-1]:
         # keep only each receiver's k nearest senders (reference uses topk then
         # masks the rest, gcbf/env/dubins_car.py:736-740, but with a Python
-        # loop per row — this is the batched equivalent)
+        # loop per row — this is the batched equivalent).  Tie semantics
+        # differ from torch.topk: EVERY sender tied at the k-th smallest
+        # distance is kept (can exceed k on exact float ties — measure-zero
+        # in practice); the HIP kth_smallest kernel matches this rule.
         kth = dist.topk(max_neighbors, dim=-1, largest=False).values[..., -1:]
         dist = torch.where(dist <= kth, dist, dist + big)
 

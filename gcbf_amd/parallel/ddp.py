@@ -84,6 +84,12 @@ class GradSynchronizer:
             grads = [p.grad for p in params if p.grad is not None]
             if not grads:
                 continue
+            # the flat layout is positional: a rank-divergent None pattern
+            # (e.g. a rank-conditional loss term) would silently mix
+            # gradients of different parameters across ranks
+            assert len(grads) == len(params), (
+                f"group {gi}: {len(params) - len(grads)} params have no "
+                f"grad; flat all-reduce layout would diverge across ranks")
             numel = sum(g.numel() for g in grads)
             buf = self._buffers[gi]
             if buf is None or buf.numel() != numel:
