@@ -42,8 +42,9 @@ import torch
 
 from gcbf_amd.algo import make_algo
 from gcbf_amd.env import make_env
-from gcbf_amd.parallel import (GradSynchronizer, broadcast_modules,
-                               cleanup_distributed, init_distributed)
+from gcbf_amd.parallel import (broadcast_modules,
+                               cleanup_distributed, init_distributed,
+                               make_grad_synchronizer)
 from gcbf_amd.trainer.utils import read_params, set_seed
 
 SCHEDULE_TOTAL = 500_000      # the paper run's exploration schedule length
@@ -100,7 +101,7 @@ def run(args):
                      batch_size=args.batch_size, hyperparams=hyper)
     if world_size > 1:
         broadcast_modules([algo.cbf, algo.actor])
-        algo.grad_sync = GradSynchronizer([algo.cbf, algo.actor])
+        algo.grad_sync = make_grad_synchronizer([algo.cbf, algo.actor])
     if dtype == "bf16":
         from gcbf_amd.utils.amp import enable_bf16
         enable_bf16(algo)

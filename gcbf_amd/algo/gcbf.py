@@ -361,8 +361,16 @@ class GCBF(Algorithm):
             acc_unsafe.detach(), acc_safe.detach(), acc_h_dot.detach()])
 
     def _update_tail(self, step, writer, logs, inner_iter) -> dict:
-        # one host sync for the whole update's scalars
-        log_vals = torch.stack(logs).cpu()
+        # one host sync for the whole update's scalars; under DP the stack
+        # is mean-reduced across ranks first, so rank-0 TensorBoard curves
+        # reflect the GLOBAL batch (one small collective per update, not
+        # one per scalar; reference semantics gcbf/algo/gcbf.py:229-237)
+        log_stack = torch.stack(logs).detach()
+        import torch.distributed as dist
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(log_stack, op=dist.ReduceOp.SUM)
+            log_stack /= dist.get_world_size()
+        log_vals = log_stack.cpu()
         if writer is not None:
             names = ("loss/unsafe", "loss/safe", "loss/derivative",
                      "loss/action", "acc/unsafe", "acc/safe", "acc/derivative")

@@ -1,5 +1,7 @@
-from .ddp import (GradSynchronizer, all_reduce_scalar, broadcast_modules,
-                  cleanup_distributed, env_world, init_distributed)
+from .ddp import (BucketedGradSynchronizer, GradSynchronizer,
+                  all_reduce_scalar, broadcast_modules, cleanup_distributed,
+                  env_world, init_distributed, make_grad_synchronizer)
 
-__all__ = ["GradSynchronizer", "all_reduce_scalar", "broadcast_modules",
-           "cleanup_distributed", "env_world", "init_distributed"]
+__all__ = ["BucketedGradSynchronizer", "GradSynchronizer",
+           "all_reduce_scalar", "broadcast_modules", "cleanup_distributed",
+           "env_world", "init_distributed", "make_grad_synchronizer"]

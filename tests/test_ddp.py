@@ -116,6 +116,113 @@ def test_gcbf_dp_two_ranks_stay_in_sync():
     assert torch.allclose(results[0], results[1], atol=1e-6)
 
 
+def _worker_bucketed(rank, world, q, port):
+    _init(rank, world, port)
+    from gcbf_amd.parallel import BucketedGradSynchronizer
+    torch.manual_seed(rank)
+    m = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(),
+                            torch.nn.Linear(32, 8), torch.nn.Linear(8, 2))
+    broadcast_modules([m])
+    # tiny bucket size forces several buckets (layout still rank-invariant)
+    sync = BucketedGradSynchronizer([m], bucket_bytes=512)
+    sync.install()
+    x = torch.randn(4, 16)
+    loss = m(x).pow(2).sum()
+    loss.backward()
+    local = [p.grad.clone() for p in m.parameters()]
+    sync()
+    q.put((rank, local, [p.grad.clone() for p in m.parameters()]))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_bucketed_grad_synchronizer_averages():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    ps = [ctx.Process(target=_worker_bucketed, args=(r, 2, q, port))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, local, synced = q.get(timeout=120)
+        results[rank] = (local, synced)
+    for p in ps:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    for g0, g1, s0, s1 in zip(results[0][0], results[1][0],
+                              results[0][1], results[1][1]):
+        mean = (g0 + g1) / 2
+        assert torch.allclose(s0, mean, atol=1e-6)
+        assert torch.allclose(s1, mean, atol=1e-6)
+
+
+class _CollectWriter:
+    def __init__(self):
+        self.scalars = []
+
+    def add_scalar(self, name, value, step):
+        self.scalars.append((name, value, step))
+
+
+def _worker_gcbf_bucketed_logs(rank, world, q, port):
+    _init(rank, world, port)
+    from gcbf_amd.algo import make_algo
+    from gcbf_amd.env import make_env
+    from gcbf_amd.parallel import make_grad_synchronizer
+    from gcbf_amd.trainer.utils import set_seed
+    set_seed(100 + rank)
+    dev = torch.device("cpu")
+    env = make_env("SimpleCar", 4, dev)
+    env.train()
+    algo = make_algo("gcbf", env, 4, env.node_dim, env.edge_dim,
+                     env.action_dim, dev, batch_size=20)
+    broadcast_modules([algo.cbf, algo.actor])
+    algo.grad_sync = make_grad_synchronizer([algo.cbf, algo.actor])
+    writer = _CollectWriter()
+    data = env.reset()
+    for step in range(1, 21):
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.5)
+        data, r, done, info = env.step(a)
+        if done:
+            data = env.reset()
+        if algo.is_update(step):
+            algo.update(step, writer)
+    w = algo.actor.feat_2_action.net[0].weight.detach().clone()
+    q.put((rank, w, writer.scalars))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_gcbf_dp_bucketed_sync_and_global_logs():
+    """Bucketed (overlapped) all-reduce keeps ranks in lockstep AND the
+    logged update scalars are all-reduced so every rank logs the same
+    global-batch values."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    ps = [ctx.Process(target=_worker_gcbf_bucketed_logs,
+                      args=(r, 2, q, port))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, w, scalars = q.get(timeout=600)
+        results[rank] = (w, scalars)
+    for p in ps:
+        p.join(timeout=600)
+        assert p.exitcode == 0
+    assert torch.allclose(results[0][0], results[1][0], atol=1e-6)
+    s0, s1 = results[0][1], results[1][1]
+    assert len(s0) == len(s1) > 0
+    for (n0, v0, t0), (n1, v1, t1) in zip(s0, s1):
+        assert n0 == n1 and t0 == t1
+        assert abs(v0 - v1) < 1e-6, (n0, v0, v1)
+
+
 def test_env_world_defaults():
     from gcbf_amd.parallel import env_world
     rank, world, local = env_world()

@@ -14,8 +14,9 @@ import torch
 
 from gcbf_amd.algo import make_algo
 from gcbf_amd.env import make_env
-from gcbf_amd.parallel import (GradSynchronizer, broadcast_modules,
-                               cleanup_distributed, init_distributed)
+from gcbf_amd.parallel import (broadcast_modules,
+                               cleanup_distributed, init_distributed,
+                               make_grad_synchronizer)
 from gcbf_amd.trainer import Trainer
 from gcbf_amd.trainer.utils import init_logger, read_params, set_seed
 
@@ -89,7 +90,7 @@ def train(args):
 
     if world_size > 1:
         broadcast_modules([algo.cbf, algo.actor])
-        algo.grad_sync = GradSynchronizer([algo.cbf, algo.actor])
+        algo.grad_sync = make_grad_synchronizer([algo.cbf, algo.actor])
 
     trainer = Trainer(env=env, env_test=env_test, algo=algo, log_dir=log_path,
                       rank=rank, world_size=world_size)
