@@ -29,8 +29,11 @@ from torch import Tensor
 
 from .. import ops
 from ..graph import GraphBatch
+from .demo_world import BoxWorld
 from .simple_car import SimpleCar
 from .utils import plot_graph, fig_to_rgb_array, rejection_sample_positions
+
+N_LIDAR_RAYS = 32  # reference gcbf/env/dubins_car.py:313
 
 
 class DubinsCar(SimpleCar):
@@ -42,6 +45,15 @@ class DubinsCar(SimpleCar):
         self._num_obs = self._params["num_obs"]
         self._params["obs_len_max"] = self._params["area_size"] / 8.0
         self._obs: Optional[Tensor] = None  # obstacle states (n_obs, 4)
+        # demo modes: native box world (analytic LiDAR + contacts — the
+        # reference used pybullet here, gcbf/env/dubins_car.py:55-382)
+        self._world: Optional[BoxWorld] = None
+
+    def demo(self, idx: int):
+        super().demo(idx)
+        if idx == 3:
+            # demo_3 uses small moving obstacles (reference :74-75)
+            self._params["obs_len_max"] = self._params["car_radius"] * 2
 
     @property
     def max_episode_steps(self) -> int:
@@ -132,10 +144,8 @@ class DubinsCar(SimpleCar):
         r = self._params["car_radius"]
         obs_r = self._params["obs_point_r"]
 
-        if self._mode not in ("train", "test", "demo_2"):
-            raise RuntimeError(
-                f"mode {self._mode}: pybullet demo modes need the optional "
-                f"pybullet dependency (not available in this build)")
+        if self._mode in ("demo_0", "demo_1", "demo_3"):
+            return self._reset_demo()
 
         # obstacles: uniform positions, random heading/speed
         # (reference gcbf/env/dubins_car.py:392-401)
@@ -177,6 +187,195 @@ class DubinsCar(SimpleCar):
             torch.cat([states[:, :2], goals[:, :2], obs_pos], dim=0))
         return self._data
 
+    # ------------------------------------------------------- demo modes
+    # Native replacements for the reference's pybullet demo paths
+    # (gcbf/env/dubins_car.py:55-382, 637-722, 884-923): oriented-box
+    # obstacles, analytic 32-ray LiDAR (occluded by other agents/goals),
+    # SDF contact checks, kinematic box motion, matplotlib rendering.
+    def _init_obs_world(self) -> BoxWorld:
+        """Obstacle layout of _init_obs_bullet (reference :147-209):
+        alternating vertical/horizontal boxes in an intersection corridor;
+        demo_3 adds 4 large static corner blocks and rejects overlapping
+        small boxes."""
+        import numpy as np
+        p = self._params
+        area = p["area_size"]
+        world = BoxWorld(self.device)
+        for i in range(self._num_obs):
+            center = np.random.rand(3) * area / 4
+            if i % 2 == 0:
+                center[0] += area / 2 - area / 8
+                center[1] *= 4
+                theta = float(np.pi / 2)
+            else:
+                center[1] += area / 2 - area / 8
+                center[0] *= 4
+                theta = 0.0
+            length = float(np.random.rand()) * p["obs_len_max"] + area / 80
+            width = area / 80
+            if self._mode == "demo_3" and world.num_boxes:
+                # reject boxes overlapping an existing one (closest points
+                # within obs_point_r, reference :177-186); the reference's
+                # for-loop does not retry rejected indices
+                probe = torch.tensor([center[:2]], dtype=torch.float32,
+                                     device=self.device)
+                # conservative surface gap: center distance minus both
+                # half-diagonals
+                if float(world.box_distance(probe).min()) < \
+                        (length + width) / 2 + p["obs_point_r"]:
+                    continue
+            speed = float(2 * np.random.rand() - 1) * p["obs_speed_limit"]
+            world.add_box(center[:2], (length, width), theta,
+                          vel=(theta, speed))
+        if self._mode == "demo_3":
+            sq = area / 16 * 3
+            for cx, cy in ((sq, sq), (sq, area - sq), (area - sq, sq),
+                           (area - sq, area - sq)):
+                world.add_box((cx, cy), (sq / 2, sq / 4), 0.0, vel=(0.0, 0.0))
+        return world
+
+    def _sample_demo_positions(self, kind: str) -> Tensor:
+        """Cross/random layout rejection sampling of _init_agent_bullet /
+        _init_goal_bullet (reference :211-313)."""
+        import numpy as np
+        p = self._params
+        side = p["area_size"]
+        r = p["car_radius"]
+        cross = self._mode in ("demo_0", "demo_3")
+        min_sep = 4 * r if kind == "agent" else 8 * r
+        clearance = 2 * r + 2 * p["obs_point_r"]
+        pos = np.zeros((self.num_agents, 2))
+        i, guard = 0, 0
+        while i < self.num_agents:
+            guard += 1
+            if guard > 100000:
+                raise RuntimeError("demo placement: no valid layout found")
+            if cross:
+                cand = np.random.rand(2) * side / 8 * 3
+                if kind == "agent":
+                    if i % 2 == 0:
+                        cand[0] += side / 16 * 13 - side / 16 * 3
+                    if self._mode == "demo_3" and i % 4 in (3, 0):
+                        cand[1] += side / 16 * 13 - side / 16 * 3
+                else:
+                    if i % 2 == 1:
+                        cand[0] += side / 16 * 13
+                    cand[1] += side / 16 * 13
+                    if self._mode == "demo_3" and i % 4 in (3, 0):
+                        cand[1] -= side / 16 * 13
+            else:  # demo_1: random placement
+                cand = np.random.rand(2) * side
+            if i and np.linalg.norm(pos[:i] - cand, axis=1).min() <= min_sep:
+                continue
+            if self._world.num_boxes:
+                probe = torch.tensor(np.array([cand]),
+                                     dtype=torch.float32,
+                                     device=self.device)
+                # surface-to-surface: box SDF minus the circle footprint
+                if float(self._world.box_distance(probe).min()) - r \
+                        <= clearance:
+                    continue
+            pos[i] = cand
+            i += 1
+        return torch.tensor(pos, dtype=torch.float32, device=self.device)
+
+    def _lidar_observe(self, agent_pos: Tensor) -> Tensor:
+        """32 rays per agent against the box world, occluded by other
+        agents and goal cylinders; hits become the obstacle point cloud
+        self._obs = [x, y, heading, speed] (reference _lidar /
+        _get_observation_bullet, :313-382).  Returns the (K, 2) points."""
+        n = self.num_agents
+        dev = self.device
+        angles = torch.arange(N_LIDAR_RAYS, device=dev) \
+            * (2 * torch.pi / N_LIDAR_RAYS)
+        dirs1 = torch.stack([torch.cos(angles), torch.sin(angles)], dim=1)
+        hits_pos, hits_vel = [], []
+        goal_pos = self._goal[:, :2]
+        for a in range(n):
+            origins = agent_pos[a].expand(N_LIDAR_RAYS, 2)
+            others = torch.cat([agent_pos[:a], agent_pos[a + 1:], goal_pos])
+            hit, pts, box = self._world.raycast(
+                origins, dirs1, self._params["comm_radius"],
+                occluder_centers=others,
+                occluder_radius=self._params["car_radius"])
+            if hit.any():
+                # reference appends in reversed ray order (:330)
+                idx = hit.nonzero(as_tuple=True)[0].flip(0)
+                hits_pos.append(pts[idx])
+                hits_vel.append(self._world.vel[box[idx]])
+        if hits_pos:
+            obs_pos = torch.cat(hits_pos)
+            obs_vel = torch.cat(hits_vel)
+        else:
+            obs_pos = torch.zeros(0, 2, device=dev)
+            obs_vel = torch.zeros(0, 2, device=dev)
+        self._obs = torch.cat([obs_pos, obs_vel], dim=1)
+        return obs_pos
+
+    def _reset_demo(self) -> GraphBatch:
+        import numpy as np
+        self._world = self._init_obs_world() \
+            if self._mode in ("demo_0", "demo_3") else BoxWorld(self.device)
+        agent_pos = self._sample_demo_positions("agent")
+        theta = torch.tensor(
+            np.arctan2(np.sin(np.random.rand(self.num_agents) * 2 * np.pi),
+                       np.cos(np.random.rand(self.num_agents) * 2 * np.pi)),
+            dtype=torch.float32, device=self.device)
+        states = torch.cat([
+            agent_pos, theta.unsqueeze(1),
+            torch.zeros(self.num_agents, 1, device=self.device)], dim=1)
+        goal_pos = self._sample_demo_positions("goal")
+        if self._mode == "demo_3":  # reference shuffles goals (:311-312)
+            goal_pos = goal_pos[torch.randperm(self.num_agents,
+                                               device=self.device)]
+        self._goal = torch.cat(
+            [goal_pos, torch.zeros(self.num_agents, 2, device=self.device)],
+            dim=1)
+        obs_pos = self._lidar_observe(agent_pos)
+        data = self._build_data(states)
+        self._data = self.add_communication_links(data)
+        self._set_plot_limits(
+            torch.cat([states[:, :2], goal_pos, obs_pos], dim=0))
+        return self._data
+
+    def _step_demo(self, action: Tensor):
+        """Demo-mode step (reference :525-615): integrate agents on the
+        current graph, advance the boxes kinematically, re-scan LiDAR,
+        rebuild the graph from scratch (node count varies with hits)."""
+        self._t += 1
+        reward_action = -torch.norm(action, dim=1).sum() * 0.01
+        action = action + self._step_u_ref()
+        lower_lim, upper_lim = self.action_lim
+        action = torch.clamp(action, lower_lim, upper_lim)
+        am = self._data.agent_mask
+        sel = am if am is not None else slice(None)
+        prev_reach = torch.less(
+            torch.norm(self.data.states[sel, :2] - self._goal[:, :2], dim=1),
+            self._params["dist2goal"])
+        with torch.no_grad():
+            state = self.forward(self._data, action)
+        agent_state = state[sel]
+
+        if self._mode in ("demo_0", "demo_3"):
+            self._world.advance(self.dt)
+            self._lidar_observe(agent_state[:, :2])
+        else:  # demo_1: no boxes; keep the (empty) point cloud
+            self._lidar_observe(agent_state[:, :2])
+        data = self._build_data(agent_state)
+        self._data = self.add_communication_links(data)
+
+        time_up = self._t >= self.max_episode_steps
+        reach = torch.less(
+            torch.norm(agent_state[:, :2] - self._goal[:, :2], dim=1),
+            self._params["dist2goal"])
+        done = bool(time_up or reach.all())
+        collision = self.collision_mask(self._data)
+        reward = (reach.int() - prev_reach.int()).int() * 10 \
+            - collision.int() * 0.1 - 0.0001 + reward_action
+        safe = 1.0 - collision.sum() / self.num_agents
+        return self.data, reward.detach(), done, {
+            "reach": reach, "collision": collision, "safe": safe}
+
     def _build_data(self, agent_states: Tensor) -> GraphBatch:
         n_obs = self._obs.shape[0]
         x = torch.cat([
@@ -193,6 +392,8 @@ class DubinsCar(SimpleCar):
 
     # ------------------------------------------------------------------ step
     def step(self, action: Tensor) -> Tuple[GraphBatch, Tensor, bool, dict]:
+        if self._mode in ("demo_0", "demo_1", "demo_3"):
+            return self._step_demo(action)
         self._t += 1
         # fused single-kernel path on GPU (ops/hip/env_step.hip)
         out = ops.env_step_fused(
@@ -370,8 +571,26 @@ class DubinsCar(SimpleCar):
 
     def collision_mask(self, data: GraphBatch) -> Tensor:
         r = self._params["car_radius"]
-        if self._mode not in ("train", "test", "demo_1", "demo_2"):
-            raise NotImplementedError
+        if self._mode in ("demo_0", "demo_3"):
+            # agent-agent circles PLUS contact with the box obstacles
+            # (reference :900-920 used pybullet getClosestPoints over the
+            # env's CURRENT bodies for every graph in a batch — mirrored:
+            # the box term uses the env's current agent positions)
+            B = data.num_graphs
+            N = data.nodes_per_graph
+            n = self.num_agents
+            sv = data.states.view(B, N, -1)[:, :n, :2]
+            pd = sv.unsqueeze(2) - sv.unsqueeze(1)
+            dist = pd.norm(dim=-1) + torch.eye(
+                n, device=data.device) * (2 * r + 1)
+            coll_agent = (dist < 2 * r).max(dim=2)[0].reshape(-1)
+            am = self._data.agent_mask
+            cur = self._data.states[am if am is not None else slice(None), :2]
+            coll_box = (self._world.min_distance(cur) - r <= 0) \
+                if self._world is not None and self._world.num_boxes \
+                else torch.zeros(n, dtype=torch.bool, device=data.device)
+            return torch.logical_or(
+                coll_agent, coll_box.repeat(B)).bool()
         m = self._fused_mask(data, "collision")
         if m is not None:
             return m
@@ -379,11 +598,68 @@ class DubinsCar(SimpleCar):
         return (dist < 2 * r).max(dim=2)[0].reshape(-1).bool()
 
     # ---------------------------------------------------------------- render
+    def _render_demo(self, plot_edge: bool = True):
+        """Top-down matplotlib frame of the demo world: boxes, agents,
+        goals, LiDAR hit points and communication edges — the functional
+        equivalent of the reference's pybullet camera image
+        (gcbf/env/dubins_car.py:637-722; documented deviation: matplotlib
+        instead of a GL render)."""
+        import matplotlib.pyplot as plt
+        from matplotlib.patches import Circle, Rectangle
+        from matplotlib.transforms import Affine2D
+        import numpy as np
+        r = self._params["car_radius"]
+        fig, ax = plt.subplots(1, 1, figsize=(12, 10), dpi=100)
+        w = self._world
+        for b in range(w.num_boxes):
+            cx, cy = w.centers[b].tolist()
+            hx, hy = w.half[b].tolist()
+            th = float(w.theta[b])
+            rect = Rectangle((cx - hx, cy - hy), 2 * hx, 2 * hy,
+                             color="#8B0000", alpha=0.9)
+            rect.set_transform(
+                Affine2D().rotate_around(cx, cy, th) + ax.transData)
+            ax.add_patch(rect)
+        data = self.data
+        am = data.agent_mask
+        states = data.states[am if am is not None else slice(None)]
+        pts = data.states[~am][:, :2].cpu().numpy() if am is not None \
+            else np.zeros((0, 2))
+        if len(pts):
+            ax.scatter(pts[:, 0], pts[:, 1], s=12, c="#EEA333", zorder=3)
+        for i in range(self.num_agents):
+            x, y, th = float(states[i, 0]), float(states[i, 1]), \
+                float(states[i, 2])
+            ax.add_patch(Circle((x, y), r, color="#FF8C00", alpha=0.8,
+                                zorder=4))
+            ax.plot([x, x + r * np.cos(th)], [y, y + r * np.sin(th)],
+                    c="k", lw=1, zorder=5)
+            gx, gy = float(self._goal[i, 0]), float(self._goal[i, 1])
+            ax.add_patch(Circle((gx, gy), r, color="#3CB371", alpha=0.8,
+                                zorder=4))
+        if plot_edge and data.num_edges:
+            ei = data.edge_index.cpu().numpy()
+            pos = data.pos.cpu().numpy()
+            for k in range(ei.shape[1]):
+                a, b = ei[0, k], ei[1, k]
+                ax.plot([pos[a, 0], pos[b, 0]], [pos[a, 1], pos[b, 1]],
+                        c="#1A5276", lw=0.5, alpha=0.5, zorder=2)
+        x_int = self._xy_max[0] - self._xy_min[0]
+        y_int = self._xy_max[1] - self._xy_min[1]
+        ax.set_xlim(self._xy_min[0], self._xy_min[0] + max(x_int, y_int))
+        ax.set_ylim(self._xy_min[1], self._xy_min[1] + max(x_int, y_int))
+        ax.set_aspect("equal")
+        plt.axis("off")
+        plt.tight_layout()
+        frame = fig_to_rgb_array(fig)
+        plt.close(fig)
+        return frame
+
     def render(self, traj=None, return_ax: bool = False, plot_edge: bool = True,
                ax=None):
         import matplotlib.pyplot as plt
-        if self._mode not in ("train", "test", "demo_2"):
-            raise RuntimeError("pybullet demo rendering unavailable")
+        if self._mode in ("demo_0", "demo_1", "demo_3") and traj is None:
+            return self._render_demo(plot_edge=plot_edge)
         return_tuple = True
         if traj is None:
             traj = (self.data,)
