@@ -259,10 +259,14 @@ class GCBF(Algorithm):
         if self._ring is not None and self._ring.usable(graph_list):
             try:
                 graphs = self._ring.batch(graph_list)
-            except Exception as e:
+            except Exception:
+                # metadata-only snapshots (ring-backed rollout) cannot fall
+                # back to from_list — there are no stored tensors to
+                # concatenate; fail loudly rather than mix gradients
+                if any(g.states is None for g in graph_list):
+                    raise
                 import warnings
-                warnings.warn(f"ring batch failed ({e}); "
-                              f"falling back to from_list")
+                warnings.warn("ring batch failed; falling back to from_list")
                 self._ring = None
         if graphs is None:
             graphs = GraphBatch.from_list(graph_list)

@@ -5,9 +5,12 @@ state [x, y, θ, v], action [ω/10, a], obstacles as graph nodes (node feature
 x=1), dense radius graph with agent-only receivers and optional k-nearest
 cap, PID-style reference controller.
 
-The pybullet demo modes (demo_0/1/3: LiDAR point clouds, URDF rendering)
-require pybullet and are gated behind its availability; train/test/demo_2
-paths are fully native.
+The demo modes (demo_0/1/3) are fully native: the reference's pybullet
+LiDAR/contact/camera machinery (gcbf/env/dubins_car.py:55-382, 637-722,
+884-923) is replaced by an analytic oriented-box world
+(gcbf_amd/env/demo_world.py) — 32-ray slab-test LiDAR with agent/goal
+occluders, SDF contact checks, kinematic box motion, matplotlib top-down
+rendering — vectorized in torch and device-aware (runs on the MI355X).
 
 Reference quirks reproduced on purpose (parity):
 * ``dynamics`` clamps the *position* speed via clamp(v, max=limit) but the

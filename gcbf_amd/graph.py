@@ -44,7 +44,7 @@ class GraphBatch:
 
     __slots__ = ("x", "pos", "states", "edge_index", "edge_attr", "agent_mask",
                  "u_ref", "_ptr", "_dst_ptr", "seg_dst", "agents_first_n",
-                 "ring_id", "agent_index")
+                 "ring_id", "agent_index", "_edge_count")
 
     def __init__(
             self,
@@ -79,6 +79,7 @@ class GraphBatch:
         # optional: slot id in the update engine's device ring (stamped by
         # Buffer.on_append when the captured update engine is active)
         self.ring_id = None
+        self._edge_count = None
         # optional: LONG index tensor of agent rows (uniform batches have a
         # static agent layout) — models prefer it over the boolean
         # agent_mask because integer indexing is hipGraph-capturable
@@ -99,11 +100,14 @@ class GraphBatch:
 
     @property
     def num_nodes(self) -> int:
-        return self.states.shape[0]
+        return (self.states if self.states is not None else self.x).shape[0]
 
     @property
     def num_edges(self) -> int:
-        return 0 if self.edge_index is None else self.edge_index.shape[1]
+        if self.edge_index is None:
+            # metadata-only snapshots (ring-backed rollout) carry the count
+            return getattr(self, "_edge_count", None) or 0
+        return self.edge_index.shape[1]
 
     @property
     def num_graphs(self) -> int:

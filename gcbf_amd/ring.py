@@ -56,6 +56,18 @@ class RingStore:
         g.ring_id = self.next_id
         self.next_id += 1
 
+    def push_raw(self, states, u_ref) -> int:
+        """Copy raw state/u_ref tensors into the next slot; returns the
+        ring id.  Used by the rollout engine to append WITHOUT cloning a
+        full graph snapshot first (the clone would be dead weight: batches
+        are rebuilt from the ring, never from the stored tensors)."""
+        slot = self.next_id % self.CAP
+        self.states[slot].copy_(states, non_blocking=True)
+        self.uref[slot].copy_(u_ref, non_blocking=True)
+        rid = self.next_id
+        self.next_id += 1
+        return rid
+
     def resident(self, ring_id: int) -> bool:
         return ring_id is not None and self.next_id - ring_id <= self.CAP
 

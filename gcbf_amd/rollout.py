@@ -97,6 +97,13 @@ class RolloutEngine:
 
         self._load_graph_from_env()
         self._capture()
+        # attach the ring store up front so per-step snapshots can be
+        # metadata-only from the first step (GCBF builds it lazily at the
+        # first update otherwise)
+        if getattr(algo, "_ring", None) is None \
+                and hasattr(algo, "_make_ring") \
+                and not getattr(algo, "_ring_tried", True):
+            algo._make_ring()
 
     # ----------------------------------------------------------------- body
     def _step_args(self, action):
@@ -227,17 +234,30 @@ class RolloutEngine:
         """One training env step.  Returns done."""
         if self._eager:
             return self._eager_step(prob)
-        # clone the CURRENT graph for the replay buffer before the replay
-        # overwrites the static buffers
+        # snapshot the CURRENT graph for the replay buffer before the replay
+        # overwrites the static buffers.  With the ring store attached the
+        # snapshot is metadata-only (2 copies into the ring instead of 5
+        # tensor clones per step — update batches are rebuilt from the ring,
+        # never from stored graph tensors; rocprof r02 measured the clones
+        # at ~25% of the rollout step)
         E = self.E
-        snap = GraphBatch(
-            x=self.x,  # static content, shared
-            pos=self.states[:, :self.pos_dim].clone(),
-            states=self.states.clone(),
-            edge_index=self.ei[:, :E].clone(),
-            edge_attr=self.ea[:E].clone(),
-            agent_mask=self.agent_mask,
-            u_ref=self.u_ref.clone())
+        ring = getattr(self.algo, "_ring", None)
+        if ring is not None and self.algo.buffer.on_append is ring.push:
+            rid = ring.push_raw(self.states, self.u_ref)
+            snap = GraphBatch(
+                x=self.x, pos=None, states=None,
+                agent_mask=self.agent_mask)
+            snap.ring_id = rid
+            snap._edge_count = E
+        else:
+            snap = GraphBatch(
+                x=self.x,  # static content, shared
+                pos=self.states[:, :self.pos_dim].clone(),
+                states=self.states.clone(),
+                edge_index=self.ei[:, :E].clone(),
+                edge_attr=self.ea[:E].clone(),
+                agent_mask=self.agent_mask,
+                u_ref=self.u_ref.clone())
 
         if np.random.rand() < prob:
             self.g_explore.replay()

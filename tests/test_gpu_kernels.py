@@ -694,7 +694,10 @@ def test_update_engine_matches_eager():
     def eager_loss(gl):
         p = algo.params
         eps, alpha = p["eps"], p["alpha"]
-        graphs = GraphBatch.from_list(gl)
+        # buffer snapshots are metadata-only under the ring-backed rollout;
+        # materialize the batch from the ring (bitwise equal to from_list —
+        # test_ring_batch_matches_from_list_gpu)
+        graphs = algo._ring.batch(gl)
         actions = algo.actor(graphs)
         graphs_next = env.forward_graph(graphs, actions)
         both = GraphBatch.from_list([graphs, graphs_next])
@@ -816,11 +819,15 @@ def test_ring_batch_matches_from_list_gpu(obs):
     env.train()
     algo = make_algo("gcbf", env, 16, env.node_dim, env.edge_dim,
                      env.action_dim, dev, batch_size=64)
-    env.reset()
-    eng = RolloutEngine(env, algo)
+    # eager stepping: the buffer keeps REAL graph tensors, so from_list is
+    # an independent ground truth for the ring rebuild
+    data = env.reset()
     for _ in range(64):
-        if eng.step(prob=0.6):
-            eng.reload()
+        data.update(u_ref=env.u_ref(data))
+        a = algo.step(data, prob=0.6)
+        data, r, done, info = env.step(a)
+        if done:
+            data = env.reset()
     algo._make_ring()
     assert algo._ring is not None
     gl = algo.buffer.sample(12, 3)

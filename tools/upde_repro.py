@@ -46,7 +46,7 @@ def eager_loss(gl):
     """Replicate GCBF._iter_eager's loss (no optimizer step)."""
     p = algo.params
     eps, alpha = p["eps"], p["alpha"]
-    graphs = GraphBatch.from_list(gl)
+    graphs = algo._ring.batch(gl)   # buffer snaps are metadata-only
     actions = algo.actor(graphs)
     graphs_next = env.forward_graph(graphs, actions)
     both = GraphBatch.from_list([graphs, graphs_next])
