@@ -85,8 +85,19 @@ class GCBF(Algorithm):
                                    edge_dim=edge_dim, phi_dim=256,
                                    action_dim=action_dim).to(device)
 
-        self.optim_cbf = Adam(self.cbf.parameters(), lr=3e-4)
-        self.optim_actor = Adam(self.actor.parameters(), lr=1e-3)
+        # fused multi-tensor Adam on GPU: one kernel per step instead of a
+        # foreach chain (~10 launches/step measured in rocprof r02);
+        # numerics identical (fp32 master weights)
+        fused = (device.type == "cuda"
+                 and os.environ.get("GCBF_AMD_FUSED_ADAM", "1") == "1")
+        try:
+            self.optim_cbf = Adam(self.cbf.parameters(), lr=3e-4,
+                                  fused=fused)
+            self.optim_actor = Adam(self.actor.parameters(), lr=1e-3,
+                                    fused=fused)
+        except (RuntimeError, ValueError):
+            self.optim_cbf = Adam(self.cbf.parameters(), lr=3e-4)
+            self.optim_actor = Adam(self.actor.parameters(), lr=1e-3)
 
         self.buffer = Buffer()   # current-episode buffer
         self.memory = Buffer()   # replay memory
@@ -237,7 +248,13 @@ class GCBF(Algorithm):
                 if prof and log7 is not None:
                     t1 = _tick(); prof["engine"] += t1 - t0; t0 = t1
             if log7 is None:
-                log7 = self._iter_eager(graph_list, prof, _tick)
+                try:
+                    log7 = self._iter_eager(graph_list, prof, _tick)
+                except Exception:
+                    if getattr(self, "_sn_ctx", None) is not None:
+                        self._sn_ctx.__exit__()
+                        self._sn_ctx = None
+                    raise
             logs.append(log7)
 
         if prof:
@@ -273,6 +290,8 @@ class GCBF(Algorithm):
         graphs.edge_attr.requires_grad_(True)
         if prof:
             t1 = _tick(); prof["batch"] += t1 - t0; t0 = t1
+        from ..nn.mlp import sn_weight_reuse
+        self._sn_ctx = sn_weight_reuse().__enter__()
         with trace_range("gcbf/forward"):
             actions = self.actor(graphs)
             # h and h_next in ONE doubled-batch CBF forward: halves the
@@ -324,6 +343,9 @@ class GCBF(Algorithm):
                 relinked = self._env.add_communication_links_batched(
                     graphs_next.detach())
                 h_next_new_link = self.cbf(relinked)
+        # σ reuse stops here: the optimizer step below changes the weights
+        self._sn_ctx.__exit__()
+        self._sn_ctx = None
         h_dot = (h_next - h) / self._env.dt
         h_dot_new_link = (h_next_new_link - h) / self._env.dt
         residue = (h_dot_new_link - h_dot).detach()

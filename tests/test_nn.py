@@ -181,3 +181,30 @@ def test_agent_index_matches_boolean_mask():
     with torch.no_grad():
         assert torch.equal(cbf(g_bool), cbf(g_idx))
         assert torch.equal(actor(g_bool), actor(g_idx))
+
+
+def test_sn_weight_reuse_context():
+    """Within sn_weight_reuse, W/σ is computed once and shared; outside,
+    every call recomputes and the power iteration advances per call."""
+    import torch
+    from gcbf_amd.nn.mlp import SNLinear, sn_weight_reuse
+    torch.manual_seed(0)
+    lin = SNLinear(8, 8)
+    lin.train()
+    with sn_weight_reuse():
+        w1 = lin.effective_weight()
+        u_after_first = lin.weight_u.clone()
+        w2 = lin.effective_weight()
+        assert w1 is w2                    # shared tensor, shared autograd
+        assert torch.equal(lin.weight_u, u_after_first)  # no 2nd advance
+    assert lin._w_cache is None            # cleared on exit
+    w3 = lin.effective_weight()
+    assert w3 is not w1
+    assert not torch.equal(lin.weight_u, u_after_first)  # advanced again
+    # gradient flows through both uses of the cached weight
+    with sn_weight_reuse():
+        x = torch.randn(4, 8)
+        y = lin(x).sum() + lin(x * 2).sum()
+        y.backward()
+    assert lin.weight_orig.grad is not None
+    assert torch.isfinite(lin.weight_orig.grad).all()
