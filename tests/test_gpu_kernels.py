@@ -839,3 +839,94 @@ def test_ring_batch_matches_from_list_gpu(obs):
     assert torch.allclose(fast.edge_attr, ref.edge_attr, atol=1e-6), \
         (fast.edge_attr - ref.edge_attr).abs().max()
     assert torch.equal(fast.u_ref, ref.u_ref)
+
+
+def test_build_graph_padded_exact_capacity():
+    """Adversarial: E_max exactly equal to the true edge count — every
+    edge present, e_count exact, no pad rows."""
+    from gcbf_amd import _C
+    from gcbf_amd import ops
+    torch.manual_seed(0)
+    dev = torch.device("cuda")
+    B, N, n_rec = 3, 12, 12
+    pos = torch.rand(B * N, 2, device=dev) * 2.0
+    states = torch.cat([pos, torch.randn(B * N, 2, device=dev)], dim=1)
+    ei_exact, ea_exact = _C.build_graph(
+        pos.contiguous(), states.contiguous(), B, n_rec, 1.0, -1,
+        int(ops.ATTR_DIFF), 4)
+    E = ei_exact.shape[1]
+    if E == 0:
+        pytest.skip("no edges for this layout")
+    ei, seg, ea, ecount = _C.build_graph_padded(
+        pos.contiguous(), states.contiguous(), B, n_rec, 1.0, -1,
+        int(ops.ATTR_DIFF), 4, E)   # capacity == exact count
+    assert int(ecount[0]) == E
+    assert torch.equal(ei[:, :E], ei_exact)
+    assert torch.allclose(ea[:E], ea_exact, atol=1e-6)
+    assert (seg[:E] < B * N).all()          # no sentinel inside real rows
+
+
+def test_build_graph_padded_overflow_is_flagged_not_oob():
+    """Adversarial: capacity BELOW the true edge count — e_count reports
+    the true count (callers detect overflow), buffers stay well-formed
+    (indices in range, attrs finite), and no out-of-bounds write occurs
+    (validated under AMD_SERIALIZE_KERNEL in the sanitizer pass)."""
+    from gcbf_amd import _C
+    from gcbf_amd import ops
+    torch.manual_seed(1)
+    dev = torch.device("cuda")
+    B, N, n_rec = 2, 16, 16
+    pos = torch.rand(B * N, 2, device=dev) * 0.5   # dense: many edges
+    states = torch.cat([pos, torch.randn(B * N, 2, device=dev)], dim=1)
+    ei_exact, _ = _C.build_graph(
+        pos.contiguous(), states.contiguous(), B, n_rec, 1.0, -1,
+        int(ops.ATTR_DIFF), 4)
+    E = ei_exact.shape[1]
+    assert E > 8, "layout should be dense"
+    E_max = E // 2
+    # guard tensors around the allocation to catch OOB writes
+    canary_lo = torch.full((256,), 7.0, device=dev)
+    ei, seg, ea, ecount = _C.build_graph_padded(
+        pos.contiguous(), states.contiguous(), B, n_rec, 1.0, -1,
+        int(ops.ATTR_DIFF), 4, E_max)
+    canary_hi = torch.full((256,), 9.0, device=dev)
+    torch.cuda.synchronize()
+    assert int(ecount[0]) == E          # true count reported -> overflow
+    assert ei.shape[1] == E_max
+    assert (ei >= 0).all() and (ei < B * N).all()
+    assert torch.isfinite(ea).all()
+    assert (seg >= 0).all() and (seg <= B * N).all()
+    assert (canary_lo == 7.0).all() and (canary_hi == 9.0).all()
+
+
+def test_kth_smallest_topk_degenerate_ties():
+    """Adversarial: ALL pairwise distances identical (regular polygon
+    center + duplicated radius) — the k-th-smallest cap must keep ties
+    without crashing, matching the eager rule (>= k edges per row kept,
+    every kept distance <= kth)."""
+    from gcbf_amd import _C
+    from gcbf_amd import ops
+    dev = torch.device("cuda")
+    n = 8
+    # all agents at distance exactly 0.5 from each other is impossible in
+    # 2D for n>3; instead: co-located pairs produce exact zero-distance
+    # ties plus identical cross distances
+    base = torch.tensor([[0.0, 0.0], [0.3, 0.0], [0.0, 0.3], [0.3, 0.3]],
+                        device=dev)
+    pos = torch.cat([base, base])       # each point duplicated (ties)
+    states = torch.cat([pos, torch.zeros(n, 2, device=dev)], dim=1)
+    k = 3
+    ei, ea = _C.build_graph(pos.contiguous(), states.contiguous(), 1, n,
+                            1.0, k, int(ops.ATTR_DIFF), 4)
+    torch.cuda.synchronize()
+    assert torch.isfinite(ea).all()
+    # every receiver keeps at least k senders; ties may exceed k
+    dst = ei[1]
+    counts = torch.bincount(dst, minlength=n)
+    assert (counts >= k).all()
+    # parity with the eager oracle (same tie semantics)
+    from gcbf_amd.ops import eager
+    ei_ref = eager.dense_radius_graph(pos, None, 1.0, k, 1)
+    have = set(map(tuple, ei.t().tolist()))
+    want = set(map(tuple, ei_ref.t().tolist()))
+    assert have == want
