@@ -173,13 +173,14 @@ class GCBF(Algorithm):
     def _make_update_engine(self):
         """Try to build the hipGraph-captured update engine (GPU only).
 
-        EXPERIMENTAL, opt-in via GCBF_AMD_UPDATE_CAPTURE=1: the captured
-        forward/losses match the eager path, but the captured backward
-        intermittently returns corrupted gradients for a layout-dependent
-        subset of parameters (pool-aliasing in graphed backward under ROCm;
-        see gcbf_amd/update_engine.py and tools/upde_repro.py for the probe
-        trail).  Until that is root-caused the default update path stays
-        eager."""
+        Opt-in via GCBF_AMD_UPDATE_CAPTURE=1.  r02 status: gradient parity
+        vs eager is VALIDATED on hardware (tests/test_gpu_kernels.py
+        update-engine tests), but the engine measures SLOWER than the
+        improved eager path (0.17-0.18 s vs 0.13-0.14 s per update) —
+        its fixed-capacity padding and duplicated actor forward cost more
+        GPU time than the launch gaps it saves (docs/ARCHITECTURE.md,
+        "Captured update engine: measured decision").  Default stays
+        eager on measurement, not correctness."""
         self._upd_engine_tried = True
         if os.environ.get("GCBF_AMD_UPDATE_CAPTURE", "0") != "1":
             return
