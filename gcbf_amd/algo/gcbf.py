@@ -379,6 +379,11 @@ class GCBF(Algorithm):
             torch.nn.utils.clip_grad_norm_(self.actor.parameters(), 1e-3)
             self.optim_cbf.step()
             self.optim_actor.step()
+            # bf16 mirror refresh must follow EVERY step: fused Adam does
+            # not bump version counters, so lazy staleness checks miss it
+            from ..nn.fused import sync_bf16_mirrors
+            sync_bf16_mirrors(self.cbf)
+            sync_bf16_mirrors(self.actor)
 
         if prof:
             t1 = _tick(); prof["opt"] += t1 - t0

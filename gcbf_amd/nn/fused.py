@@ -26,10 +26,14 @@ from torch import Tensor
 
 ACT_NONE, ACT_RELU, ACT_TANH = 0, 1, 2
 
-# below this many rows the 128x128-tile kernel cannot fill 256 CUs and
-# hipBLASLt's small-M kernels win (measured: rollout at M=256 is ~35% slower
-# on the tile kernel; update batches at M>=8K are faster on it)
-FUSED_MIN_M = int(os.environ.get("GCBF_AMD_FUSED_MIN_M", "1024"))
+# Measured dispatch (tools/smallm_bench.py on MI355X, r02): the MFMA tile
+# kernel WINS at small M — rollout shapes M<=256 are up to 4x faster than
+# hipBLASLt (0.0085 vs 0.034 ms at M=256,K=64,N=2048) — and LOSES at
+# M>=512 for the fat K=2048 update GEMMs (0.110 vs 0.136 ms at M=12288),
+# where hipBLASLt's tuned kernels take over.  So small M routes to the
+# hand-written kernel and large M to the library (the r01 default was the
+# reverse, based on a pre-ring measurement; profiles/r02_smallm.json).
+FUSED_MAX_M = int(os.environ.get("GCBF_AMD_FUSED_MAX_M", "256"))
 
 
 def _ext():
@@ -43,18 +47,25 @@ def _pad_k_x(x: Tensor, K64: int) -> Tensor:
 
 def shapes_ok(m_padded: int, n: int) -> bool:
     return (m_padded % 128 == 0 and n % 128 == 0 and n >= 128
-            and m_padded >= FUSED_MIN_M)
+            and m_padded <= FUSED_MAX_M)
 
 
 # ---------------------------------------------------------------- mirrors
 
-def _mirror(lin):
+def _mirror(lin, force: bool = False):
     """(bf16 K-padded weight mirror, bf16 bias mirror, fp32 bias) for a
-    plain nn.Linear, refreshed when the master weight version changes."""
+    plain nn.Linear, refreshed when the master weight version changes.
+
+    ``force`` refreshes unconditionally: fused Adam
+    (torch._fused_adam_) updates parameters WITHOUT bumping their
+    version counters (measured on ROCm 7 / torch 2.10 — caught by the
+    engine-vs-eager gradient parity test), so version-keyed staleness
+    detection misses optimizer steps.  Every post-optimizer refresh path
+    must pass force=True."""
     ver = lin.weight._version + (0 if lin.bias is None
                                  else lin.bias._version)
     cache = getattr(lin, "_bf16_mirror", None)
-    if cache is not None and cache[0] == ver:
+    if not force and cache is not None and cache[0] == ver:
         return cache[1], cache[2], cache[3]
     K = lin.weight.shape[1]
     K64 = (K + 63) // 64 * 64
@@ -81,10 +92,11 @@ def _mirror(lin):
 
 def sync_bf16_mirrors(module: torch.nn.Module):
     """Refresh every existing mirror after an optimizer step (captured
-    replays skip Python, so the refresh must be explicit)."""
+    replays skip Python, so the refresh must be explicit; force=True
+    because fused Adam does not bump version counters)."""
     for m in module.modules():
         if getattr(m, "_bf16_mirror", None) is not None:
-            _mirror(m)
+            _mirror(m, force=True)
 
 
 # ----------------------------------------------------------- autograd path
