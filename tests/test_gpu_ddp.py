@@ -27,6 +27,15 @@ def _free_port() -> int:
 
 
 def _worker(rank, world, q, port):
+    try:
+        _worker_body(rank, world, q, port)
+    except Exception:
+        import traceback
+        q.put((rank, "ERROR:" + traceback.format_exc()))
+        raise
+
+
+def _worker_body(rank, world, q, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -74,6 +83,7 @@ def test_two_rank_dp_on_one_gpu():
     results = {}
     for _ in range(2):
         rank, w = q.get(timeout=600)
+        assert not (isinstance(w, str) and w.startswith("ERROR:")), w
         results[rank] = w
     for p in ps:
         p.join(timeout=600)
