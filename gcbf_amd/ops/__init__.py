@@ -242,11 +242,13 @@ def pairwise_dist_masked(states: Tensor, agent_mask: Optional[Tensor],
                                       diag_offset)
 
 
-def env_step_fused(kind: str, *args):
+def env_step_fused(kind: str, *args, out=None):
     """Fused single-graph rollout step (env_step.hip); None on CPU/no-ext.
 
     kind: "dubins" | "car" | "drone"; args are forwarded to the binding.
-    Returns (new_states, u_ref_next, reward, reach, collision).
+    Returns (new_states, u_ref_next, reward, reach, collision).  ``out``
+    optionally provides those five output buffers (ping-pong capture: the
+    kernel writes the next phase's inputs directly).
     """
     states = args[0]
     if not states.is_cuda:
@@ -256,7 +258,8 @@ def env_step_fused(kind: str, *args):
         return None
     fn = {"dubins": ext.dubins_step, "car": ext.car_step,
           "drone": ext.drone_step}[kind]
-    return fn(*[a.contiguous() if torch.is_tensor(a) else a for a in args])
+    return fn(*[a.contiguous() if torch.is_tensor(a) else a for a in args],
+              out=out)
 
 
 # env kinds understood by the fused mask kernel (mirrors masks.hip)

@@ -201,13 +201,11 @@ std::vector<torch::Tensor> fused_masks(torch::Tensor states, int64_t B,
     return {t_safe, t_uns, t_coll};
 }
 
-std::vector<torch::Tensor> build_graph_padded(torch::Tensor pos,
-                                              torch::Tensor states,
-                                              int64_t B, int64_t n_rec,
-                                              double r, int64_t topk,
-                                              int64_t attr_kind,
-                                              int64_t attr_dim,
-                                              int64_t E_max) {
+std::vector<torch::Tensor> build_graph_padded(
+        torch::Tensor pos, torch::Tensor states, int64_t B, int64_t n_rec,
+        double r, int64_t topk, int64_t attr_kind, int64_t attr_dim,
+        int64_t E_max,
+        c10::optional<std::vector<torch::Tensor>> out_opt = c10::nullopt) {
     // capture-safe variant: fixed E_max edge buffers, no host sync.
     CHECK_IN(pos);
     CHECK_IN(states);
@@ -221,11 +219,24 @@ std::vector<torch::Tensor> build_graph_padded(torch::Tensor pos,
                         (int)topk, stream);
     auto incl = counts.cumsum(0, torch::kInt32);
     auto offsets = (incl - counts).contiguous();
-    auto edge_index = torch::empty({2, E_max},
-                                   pos.options().dtype(torch::kInt64));
-    auto seg = torch::empty({E_max}, pos.options().dtype(torch::kInt64));
-    auto edge_attr = torch::empty({E_max, attr_dim}, pos.options());
-    auto e_count = torch::empty({1}, pos.options().dtype(torch::kInt32));
+    torch::Tensor edge_index, seg, edge_attr, e_count;
+    if (out_opt.has_value()) {
+        TORCH_CHECK(out_opt->size() == 4,
+                    "out must be {edge_index, seg, edge_attr, e_count}");
+        for (const auto& t : *out_opt) { CHECK_IN(t); }
+        edge_index = (*out_opt)[0];
+        seg = (*out_opt)[1];
+        edge_attr = (*out_opt)[2];
+        e_count = (*out_opt)[3];
+        TORCH_CHECK(edge_index.size(1) == E_max && seg.size(0) == E_max
+                    && edge_attr.size(0) == E_max, "E_max mismatch");
+    } else {
+        edge_index = torch::empty({2, E_max},
+                                  pos.options().dtype(torch::kInt64));
+        seg = torch::empty({E_max}, pos.options().dtype(torch::kInt64));
+        edge_attr = torch::empty({E_max, attr_dim}, pos.options());
+        e_count = torch::empty({1}, pos.options().dtype(torch::kInt32));
+    }
     launch_radius_fill(pos.data_ptr<float>(), states.data_ptr<float>(),
                        offsets.data_ptr<int>(),
                        edge_index.data_ptr<int64_t>(),
@@ -274,8 +285,17 @@ struct StepOut {
     torch::Tensor new_states, u_ref_next, reward, reach, collision;
 };
 
-static std::vector<torch::Tensor> alloc_step_out(torch::Tensor states,
-                                                 int64_t n, int64_t A) {
+static std::vector<torch::Tensor> alloc_step_out(
+        torch::Tensor states, int64_t n, int64_t A,
+        const c10::optional<std::vector<torch::Tensor>>& out) {
+    if (out.has_value()) {
+        // caller-provided output buffers (ping-pong rollout capture: the
+        // kernels write the next phase's inputs directly, no copy-back)
+        TORCH_CHECK(out->size() == 5, "out must be "
+                    "{new_states, u_ref_next, reward, reach, collision}");
+        for (const auto& t : *out) { CHECK_IN(t); }
+        return *out;
+    }
     auto f = states.options();
     auto b = states.options().dtype(torch::kBool);
     return {torch::empty_like(states), torch::empty({n, A}, f),
@@ -283,14 +303,13 @@ static std::vector<torch::Tensor> alloc_step_out(torch::Tensor states,
             torch::empty({n}, b)};
 }
 
-std::vector<torch::Tensor> dubins_step(torch::Tensor states,
-                                       torch::Tensor goal,
-                                       torch::Tensor action, double dt,
-                                       double r, double sl, double d2g,
-                                       double act_lim) {
+std::vector<torch::Tensor> dubins_step(
+        torch::Tensor states, torch::Tensor goal, torch::Tensor action,
+        double dt, double r, double sl, double d2g, double act_lim,
+        c10::optional<std::vector<torch::Tensor>> out_opt = c10::nullopt) {
     CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action);
     const int64_t N = states.size(0), n = action.size(0);
-    auto out = alloc_step_out(states, n, 2);
+    auto out = alloc_step_out(states, n, 2, out_opt);
     launch_dubins_step(states.data_ptr<float>(), goal.data_ptr<float>(),
                        action.data_ptr<float>(), out[0].data_ptr<float>(),
                        out[1].data_ptr<float>(), out[2].data_ptr<float>(),
@@ -300,13 +319,14 @@ std::vector<torch::Tensor> dubins_step(torch::Tensor states,
     return out;
 }
 
-std::vector<torch::Tensor> car_step(torch::Tensor states, torch::Tensor goal,
-                                    torch::Tensor action, torch::Tensor K,
-                                    double dt, double r, double sl,
-                                    double d2g, double act_lim) {
+std::vector<torch::Tensor> car_step(
+        torch::Tensor states, torch::Tensor goal, torch::Tensor action,
+        torch::Tensor K, double dt, double r, double sl, double d2g,
+        double act_lim,
+        c10::optional<std::vector<torch::Tensor>> out_opt = c10::nullopt) {
     CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action); CHECK_IN(K);
     const int64_t N = states.size(0);
-    auto out = alloc_step_out(states, N, 2);
+    auto out = alloc_step_out(states, N, 2, out_opt);
     launch_car_step(states.data_ptr<float>(), goal.data_ptr<float>(),
                     action.data_ptr<float>(), K.data_ptr<float>(),
                     out[0].data_ptr<float>(), out[1].data_ptr<float>(),
@@ -316,14 +336,14 @@ std::vector<torch::Tensor> car_step(torch::Tensor states, torch::Tensor goal,
     return out;
 }
 
-std::vector<torch::Tensor> drone_step(torch::Tensor states,
-                                      torch::Tensor goal,
-                                      torch::Tensor action, torch::Tensor K,
-                                      double dt, double r, double sl,
-                                      double d2g, double act_lim) {
+std::vector<torch::Tensor> drone_step(
+        torch::Tensor states, torch::Tensor goal, torch::Tensor action,
+        torch::Tensor K, double dt, double r, double sl, double d2g,
+        double act_lim,
+        c10::optional<std::vector<torch::Tensor>> out_opt = c10::nullopt) {
     CHECK_IN(states); CHECK_IN(goal); CHECK_IN(action); CHECK_IN(K);
     const int64_t N = states.size(0), n = action.size(0);
-    auto out = alloc_step_out(states, n, 3);
+    auto out = alloc_step_out(states, n, 3, out_opt);
     launch_drone_step(states.data_ptr<float>(), goal.data_ptr<float>(),
                       action.data_ptr<float>(), K.data_ptr<float>(),
                       out[0].data_ptr<float>(), out[1].data_ptr<float>(),
@@ -339,9 +359,18 @@ std::vector<torch::Tensor> drone_step(torch::Tensor states,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_masks", &fused_masks,
           "batched safe/unsafe/collision agent masks in one pass");
-    m.def("dubins_step", &dubins_step, "fused DubinsCar rollout step");
-    m.def("car_step", &car_step, "fused SimpleCar rollout step");
-    m.def("drone_step", &drone_step, "fused SimpleDrone rollout step");
+    m.def("dubins_step", &dubins_step, "fused DubinsCar rollout step",
+          py::arg("states"), py::arg("goal"), py::arg("action"),
+          py::arg("dt"), py::arg("r"), py::arg("sl"), py::arg("d2g"),
+          py::arg("act_lim"), py::arg("out") = py::none());
+    m.def("car_step", &car_step, "fused SimpleCar rollout step",
+          py::arg("states"), py::arg("goal"), py::arg("action"),
+          py::arg("K"), py::arg("dt"), py::arg("r"), py::arg("sl"),
+          py::arg("d2g"), py::arg("act_lim"), py::arg("out") = py::none());
+    m.def("drone_step", &drone_step, "fused SimpleDrone rollout step",
+          py::arg("states"), py::arg("goal"), py::arg("action"),
+          py::arg("K"), py::arg("dt"), py::arg("r"), py::arg("sl"),
+          py::arg("d2g"), py::arg("act_lim"), py::arg("out") = py::none());
     m.def("segment_attn_fwd", &segment_attn_fwd,
           "fused scatter-softmax + weighted scatter-sum (forward)");
     m.def("segment_attn_bwd", &segment_attn_bwd,
@@ -353,7 +382,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("build_graph", &build_graph,
           "batched dense radius graph + edge_attr (count/scan/fill)");
     m.def("build_graph_padded", &build_graph_padded,
-          "capture-safe radius graph into fixed E_max buffers");
+          "capture-safe radius graph into fixed E_max buffers",
+          py::arg("pos"), py::arg("states"), py::arg("B"), py::arg("n_rec"),
+          py::arg("r"), py::arg("topk"), py::arg("attr_kind"),
+          py::arg("attr_dim"), py::arg("E_max"),
+          py::arg("out") = py::none());
     m.def("fused_linear", &fused_linear,
           "MFMA bf16 GEMM with fused bias + activation epilogue");
 }

@@ -78,18 +78,31 @@ class RolloutEngine:
         self.kind = kind_by_class[type(env).__name__]
 
         dev = self.device
-        self.states = data.states.clone().contiguous()
+        # ping-pong buffer PAIRS: each captured replay reads set i and the
+        # kernels write set 1-i directly (out= bindings) — no copy-back
+        # inside the graph (the copies were ~25% of the replay, rocprof r02)
+        self.states = [data.states.clone().contiguous(),
+                       data.states.clone().contiguous()]
         self.x = data.x.clone()
         self.agent_mask = (None if data.agent_mask is None
                            else data.agent_mask.clone())
         self.goal = env._goal.clone().contiguous()
-        self.u_ref = env.u_ref(data).clone().contiguous()
-        self.ei = torch.zeros(2, self.E_max, dtype=torch.long, device=dev)
-        self.seg = torch.full((self.E_max,), self.N, dtype=torch.long,
-                              device=dev)
-        self.ea = torch.zeros(self.E_max, self.edge_dim, device=dev)
+        u0 = env.u_ref(data).clone().contiguous()
+        self.u_ref = [u0, u0.clone()]
+        self.ei = [torch.zeros(2, self.E_max, dtype=torch.long, device=dev)
+                   for _ in range(2)]
+        self.seg = [torch.full((self.E_max,), self.N, dtype=torch.long,
+                               device=dev) for _ in range(2)]
+        self.ea = [torch.zeros(self.E_max, self.edge_dim, device=dev)
+                   for _ in range(2)]
+        self.cur = 0
         self.zero_action = torch.zeros(self.n, env.action_dim, device=dev)
         self.flags = torch.zeros(3, dtype=torch.int32, device=dev)
+        # persistent step outputs (pure outputs, not ping-ponged)
+        self.reward_buf = torch.zeros(self.n, device=dev)
+        self.reach_buf = torch.zeros(self.n, dtype=torch.bool, device=dev)
+        self.coll_buf = torch.zeros(self.n, dtype=torch.bool, device=dev)
+        self.ecount_buf = torch.zeros(1, dtype=torch.int32, device=dev)
         self.E = 0
 
         if self.kind in ("car", "drone"):
@@ -106,22 +119,23 @@ class RolloutEngine:
             algo._make_ring()
 
     # ----------------------------------------------------------------- body
-    def _step_args(self, action):
+    def _step_args(self, action, i: int):
         p = self.env.params
         r_key = "drone_radius" if self.kind == "drone" else "car_radius"
         act_lim = {"car": 10.0, "dubins": 2.0, "drone": 10.0}[self.kind]
-        base = [self.states, self.goal, action]
+        base = [self.states[i], self.goal, action]
         if self.kind in ("car", "drone"):
             base.append(self.K)
         return base + [self.env.dt, p[r_key], p["speed_limit"],
                        p["dist2goal"], act_lim]
 
-    def _body(self, explore: bool = False):
-        data = GraphBatch(x=self.x, pos=self.states[:, :self.pos_dim],
-                          states=self.states, edge_index=self.ei,
-                          edge_attr=self.ea, agent_mask=self.agent_mask,
-                          u_ref=self.u_ref)
-        data.seg_dst = self.seg
+    def _body(self, explore: bool, i: int):
+        o = 1 - i
+        data = GraphBatch(x=self.x, pos=self.states[i][:, :self.pos_dim],
+                          states=self.states[i], edge_index=self.ei[i],
+                          edge_attr=self.ea[i], agent_mask=self.agent_mask,
+                          u_ref=self.u_ref[i])
+        data.seg_dst = self.seg[i]
         data.agents_first_n = self.n if self.agent_mask is not None else None
         with torch.no_grad():
             if explore:
@@ -133,64 +147,74 @@ class RolloutEngine:
             else:
                 action = self.algo.actor(data)
             unsafe_any = self.env.unsafe_mask(data).any()
-            out = ops.env_step_fused(self.kind, *self._step_args(action))
-            new_states, u_ref_next, reward, reach, collision = out
-            ei, seg, ea, ecount = self._ext.build_graph_padded(
-                new_states[:, :self.pos_dim].contiguous(), new_states, 1,
+            # kernels write set o directly (out= buffers), so the replay
+            # has no copy-back; ecount lands straight in flags[0]
+            ops.env_step_fused(
+                self.kind, *self._step_args(action, i),
+                out=[self.states[o], self.u_ref[o], self.reward_buf,
+                     self.reach_buf, self.coll_buf])
+            self._ext.build_graph_padded(
+                self.states[o][:, :self.pos_dim].contiguous(),
+                self.states[o], 1,
                 self.n if self.agent_mask is not None else self.N,
                 self.env.params["comm_radius"], -1,
-                self.env._attr_kind, self.edge_dim, self.E_max)
-            # publish flags, then advance the static buffers
-            self.flags[0].copy_(ecount[0])
-            self.flags[1].copy_(reach.all().to(torch.int32))
+                self.env._attr_kind, self.edge_dim, self.E_max,
+                out=[self.ei[o], self.seg[o], self.ea[o], self.flags[0:1]])
+            self.flags[1].copy_(self.reach_buf.all().to(torch.int32))
             self.flags[2].copy_(unsafe_any.to(torch.int32))
-            self.states.copy_(new_states)
-            self.u_ref.copy_(u_ref_next)
-            self.ei.copy_(ei)
-            self.seg.copy_(seg)
-            self.ea.copy_(ea)
 
     def _capture(self):
-        saved = (self.states.clone(), self.u_ref.clone(), self.ei.clone(),
-                 self.seg.clone(), self.ea.clone())
+        saved = (self.states[0].clone(), self.u_ref[0].clone(),
+                 self.ei[0].clone(), self.seg[0].clone(),
+                 self.ea[0].clone())
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            for _ in range(3):
-                self._body(False)
-            self._body(True)
+            for k in range(4):
+                self._body(False, k % 2)
+            self._body(True, 0)
+            self._body(True, 1)
         torch.cuda.current_stream().wait_stream(s)
-        # two graphs over the same buffers: policy steps run the actor,
-        # exploration steps (zeroed action) skip it entirely
-        self.g_policy = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_policy):
-            self._body(False)
-        self.g_explore = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_explore, pool=self.g_policy.pool()):
-            self._body(True)
-        # restore pre-warmup state
-        self.states.copy_(saved[0])
-        self.u_ref.copy_(saved[1])
-        self.ei.copy_(saved[2])
-        self.seg.copy_(saved[3])
-        self.ea.copy_(saved[4])
+        # four graphs over the paired buffers: policy/explore × phase;
+        # exploration graphs skip the actor entirely
+        self.g = {}
+        pool = None
+        for explore in (False, True):
+            for i in (0, 1):
+                g = torch.cuda.CUDAGraph()
+                if pool is None:
+                    with torch.cuda.graph(g):
+                        self._body(explore, i)
+                    pool = g.pool()
+                else:
+                    with torch.cuda.graph(g, pool=pool):
+                        self._body(explore, i)
+                self.g[(explore, i)] = g
+        # restore pre-warmup state into the current (phase-0) set
+        self.cur = 0
+        self.states[0].copy_(saved[0])
+        self.u_ref[0].copy_(saved[1])
+        self.ei[0].copy_(saved[2])
+        self.seg[0].copy_(saved[3])
+        self.ea[0].copy_(saved[4])
 
     # ---------------------------------------------------------- env plumbing
     def _load_graph_from_env(self):
-        """Refresh static buffers from the env's current graph (after
+        """Refresh the CURRENT buffer set from the env's graph (after
         reset)."""
         data = self.env.data
-        self.states.copy_(data.states)
+        c = self.cur
+        self.states[c].copy_(data.states)
         self.goal.copy_(self.env._goal)
-        self.u_ref.copy_(self.env.u_ref(data))
+        self.u_ref[c].copy_(self.env.u_ref(data))
         ei, seg, ea, ecount = self._ext.build_graph_padded(
             data.pos.contiguous(), data.states.contiguous(), 1,
             self.n if self.agent_mask is not None else self.N,
             self.env.params["comm_radius"], -1,
             self.env._attr_kind, self.edge_dim, self.E_max)
-        self.ei.copy_(ei)
-        self.seg.copy_(seg)
-        self.ea.copy_(ea)
+        self.ei[c].copy_(ei)
+        self.seg[c].copy_(seg)
+        self.ea[c].copy_(ea)
         self.E = int(ecount.item())
         self._eager = self.E > self.E_max
         if self._eager:
@@ -204,14 +228,15 @@ class RolloutEngine:
     def _sync_env_from_buffers(self):
         """Materialize the env's graph from the engine state (exact edge
         build) when entering eager fallback."""
+        c = self.cur
         data = GraphBatch(x=self.x,
-                          pos=self.states[:, :self.pos_dim].clone(),
-                          states=self.states.clone(),
+                          pos=self.states[c][:, :self.pos_dim].clone(),
+                          states=self.states[c].clone(),
                           agent_mask=self.agent_mask)
         if self.agent_mask is not None:
             data.agents_first_n = self.n
         self.env._data = self.env.add_communication_links(data)
-        self.env._data.u_ref = self.u_ref.clone()
+        self.env._data.u_ref = self.u_ref[c].clone()
 
     def _eager_step(self, prob: float) -> bool:
         """Plain (non-captured) training step, used while the scene's edge
@@ -241,9 +266,10 @@ class RolloutEngine:
         # never from stored graph tensors; rocprof r02 measured the clones
         # at ~25% of the rollout step)
         E = self.E
+        c = self.cur
         ring = getattr(self.algo, "_ring", None)
         if ring is not None and self.algo.buffer.on_append is ring.push:
-            rid = ring.push_raw(self.states, self.u_ref)
+            rid = ring.push_raw(self.states[c], self.u_ref[c])
             snap = GraphBatch(
                 x=self.x, pos=None, states=None,
                 agent_mask=self.agent_mask)
@@ -252,17 +278,15 @@ class RolloutEngine:
         else:
             snap = GraphBatch(
                 x=self.x,  # static content, shared
-                pos=self.states[:, :self.pos_dim].clone(),
-                states=self.states.clone(),
-                edge_index=self.ei[:, :E].clone(),
-                edge_attr=self.ea[:E].clone(),
+                pos=self.states[c][:, :self.pos_dim].clone(),
+                states=self.states[c].clone(),
+                edge_index=self.ei[c][:, :E].clone(),
+                edge_attr=self.ea[c][:E].clone(),
                 agent_mask=self.agent_mask,
-                u_ref=self.u_ref.clone())
+                u_ref=self.u_ref[c].clone())
 
-        if np.random.rand() < prob:
-            self.g_explore.replay()
-        else:
-            self.g_policy.replay()
+        self.g[(np.random.rand() < prob, c)].replay()
+        self.cur = c = 1 - c
 
         flags = self.flags.cpu()  # ONE host sync per step
         self.E = int(flags[0])
@@ -288,8 +312,9 @@ class RolloutEngine:
             # shares the engine's buffers; env methods are not used for
             # captured stepping)
             self.env._data = GraphBatch(
-                x=self.x, pos=self.states[:, :self.pos_dim],
-                states=self.states, edge_index=self.ei, edge_attr=self.ea,
-                agent_mask=self.agent_mask, u_ref=self.u_ref)
-            self.env._data.seg_dst = self.seg
+                x=self.x, pos=self.states[c][:, :self.pos_dim],
+                states=self.states[c], edge_index=self.ei[c],
+                edge_attr=self.ea[c],
+                agent_mask=self.agent_mask, u_ref=self.u_ref[c])
+            self.env._data.seg_dst = self.seg[c]
         return done

@@ -370,7 +370,7 @@ def test_rollout_engine_matches_eager_loop():
     states_b = []
     for step in range(len(states_a)):
         done = eng.step(prob=0.5)
-        states_b.append(eng.states.clone())
+        states_b.append(eng.states[eng.cur].clone())
         if done:
             break
 
@@ -632,9 +632,9 @@ def test_rollout_engine_soft_capacity_overflow_fallback():
         saw_eager |= was_eager
         saw_captured |= not was_eager
         done = eng.step(prob=0.5)
-        assert torch.isfinite(eng.states).all()
+        assert torch.isfinite(eng.states[eng.cur]).all()
         if not eng._eager:
-            assert torch.isfinite(eng.ea[: eng.E]).all()
+            assert torch.isfinite(eng.ea[eng.cur][: eng.E]).all()
         if done:
             eng.reload()
     assert algo.buffer.size == 120

@@ -87,6 +87,30 @@ def run_graphed():
     return e._graphed(e._nodes, e._uref, e._ei, e._ea, e._seg, e._h_new,
                       e.w_dev)
 
+# ---- probe 0: FRONT outputs vs ring-materialized batch (same sample)
+e._fill_inputs(gl)
+e.gFront.replay()
+mat = algo._ring.batch(gl)
+L = len(gl)
+nA = L * e.n
+print("uref maxdiff  :", (e._uref[:nA] - mat.u_ref).abs().max().item())
+print("nodes maxdiff :",
+      (e._nodes[:L * e.N] - mat.states).abs().max().item())
+with torch.no_grad():
+    from gcbf_amd.graph import GraphBatch
+    gchk = GraphBatch(x=e.x_tile, pos=e._nodes[:, :e.pd], states=e._nodes,
+                      edge_index=e._ei, edge_attr=e._ea,
+                      agent_mask=e.agent_mask_tile, u_ref=e._uref,
+                      ptr=e.ptr)
+    gchk.seg_dst = e._seg
+    gchk.agent_index = e.agent_index
+    a_front = algo.actor(gchk)          # eager actor on FRONT buffers
+    a_mat = algo.actor(mat)             # eager actor on ring batch
+print("eager-actor(front bufs) vs eager-actor(ring batch) maxdiff:",
+      (a_front[:nA] - a_mat).abs().max().item())
+print("mean |a|^2 front:", a_front[:nA].pow(2).sum(1).mean().item(),
+      " ring:", a_mat.pow(2).sum(1).mean().item())
+
 # ---- probe 1: replayed grads vs eager grads on the SAME batch
 e._fill_inputs(gl)
 lcap, log_cap = run_graphed()
