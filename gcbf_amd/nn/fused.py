@@ -26,14 +26,20 @@ from torch import Tensor
 
 ACT_NONE, ACT_RELU, ACT_TANH = 0, 1, 2
 
-# Measured dispatch (tools/smallm_bench.py on MI355X, r02): the MFMA tile
-# kernel WINS at small M — rollout shapes M<=256 are up to 4x faster than
-# hipBLASLt (0.0085 vs 0.034 ms at M=256,K=64,N=2048) — and LOSES at
-# M>=512 for the fat K=2048 update GEMMs (0.110 vs 0.136 ms at M=12288),
-# where hipBLASLt's tuned kernels take over.  So small M routes to the
-# hand-written kernel and large M to the library (the r01 default was the
-# reverse, based on a pre-ring measurement; profiles/r02_smallm.json).
-FUSED_MAX_M = int(os.environ.get("GCBF_AMD_FUSED_MAX_M", "256"))
+# Measured dispatch history (profiles/r02_smallm.json + rollout_bench):
+# * standalone microbenches (back-to-back calls, weights L2-resident) say
+#   the MFMA tile kernel wins at M<=256 by up to 4x and loses at M>=512;
+# * IN CONTEXT the library wins everywhere we dispatch: captured rollout
+#   policy path 0.399 ms/step with the kernel off vs 0.532 ms on
+#   (weights stream from HBM each step — the microbench advantage was a
+#   cache artifact), and the update at M~12k measures faster on
+#   hipBLASLt too (update 0.123 s vs 0.146 s per 512 steps).
+# Default is therefore library GEMMs everywhere (FUSED_MAX_M=0); the
+# hand-written kernel stays built, numerics-tested and dispatchable via
+# GCBF_AMD_FUSED_MAX_M for future shapes.  The always-on hand-written
+# HIP surface is the graph builder / env step / mask / segment-attention
+# kernel family, where no library covers the fusion.
+FUSED_MAX_M = int(os.environ.get("GCBF_AMD_FUSED_MAX_M", "0"))
 
 
 def _ext():
