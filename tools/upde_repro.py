@@ -3,6 +3,9 @@
     PYTHONPATH=. python tools/upde_repro.py
 """
 import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
