@@ -30,6 +30,38 @@ def _init(rank, world, port):
     dist.init_process_group("gloo", rank=rank, world_size=world)
 
 
+
+def _spawn_collect(worker, world=2, timeout=600, tries=2):
+    """Launch `world` spawn processes and collect one queue item each.
+
+    Retries once on OS-level spawn/queue races (FileNotFoundError /
+    ConnectionResetError observed only under heavy CPU contention)."""
+    last = None
+    for _ in range(tries):
+        try:
+            ctx = mp.get_context("spawn")
+            q = ctx.Queue()
+            port = _free_port()
+            ps = [ctx.Process(target=worker, args=(r, world, q, port))
+                  for r in range(world)]
+            for p in ps:
+                p.start()
+            results = {}
+            for _ in range(world):
+                item = q.get(timeout=timeout)
+                results[item[0]] = item[1:]
+            for p in ps:
+                p.join(timeout=timeout)
+                assert p.exitcode == 0
+            return results
+        except (FileNotFoundError, ConnectionResetError) as e:
+            last = e
+            for p in ps:
+                if p.is_alive():
+                    p.terminate()
+    raise last
+
+
 def _worker_grad_sync(rank, world, q, port):
     _init(rank, world, port)
     torch.manual_seed(rank)  # different grads per rank
@@ -48,20 +80,7 @@ def _worker_grad_sync(rank, world, q, port):
 
 
 def test_grad_synchronizer_averages():
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = _free_port()
-    ps = [ctx.Process(target=_worker_grad_sync, args=(r, 2, q, port))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    results = {}
-    for _ in range(2):
-        rank, local, synced = q.get(timeout=120)
-        results[rank] = (local, synced)
-    for p in ps:
-        p.join(timeout=120)
-        assert p.exitcode == 0
+    results = _spawn_collect(_worker_grad_sync, timeout=120)
     mean = (results[0][0] + results[1][0]) / 2
     assert torch.allclose(results[0][1], mean, atol=1e-6)
     assert torch.allclose(results[1][1], mean, atol=1e-6)
@@ -97,23 +116,9 @@ def _worker_gcbf_dp(rank, world, q, port):
 
 
 def test_gcbf_dp_two_ranks_stay_in_sync():
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = _free_port()
-    ps = [ctx.Process(target=_worker_gcbf_dp,
-                      args=(r, 2, q, port))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    results = {}
-    for _ in range(2):
-        rank, w = q.get(timeout=600)
-        results[rank] = w
-    for p in ps:
-        p.join(timeout=600)
-        assert p.exitcode == 0
+    results = _spawn_collect(_worker_gcbf_dp)
     # identical optimizer trajectories (same averaged grads every step)
-    assert torch.allclose(results[0], results[1], atol=1e-6)
+    assert torch.allclose(results[0][0], results[1][0], atol=1e-6)
 
 
 def _worker_bucketed(rank, world, q, port):
@@ -137,20 +142,7 @@ def _worker_bucketed(rank, world, q, port):
 
 
 def test_bucketed_grad_synchronizer_averages():
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = _free_port()
-    ps = [ctx.Process(target=_worker_bucketed, args=(r, 2, q, port))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    results = {}
-    for _ in range(2):
-        rank, local, synced = q.get(timeout=120)
-        results[rank] = (local, synced)
-    for p in ps:
-        p.join(timeout=120)
-        assert p.exitcode == 0
+    results = _spawn_collect(_worker_bucketed, timeout=120)
     for g0, g1, s0, s1 in zip(results[0][0], results[1][0],
                               results[0][1], results[1][1]):
         mean = (g0 + g1) / 2
@@ -200,21 +192,7 @@ def test_gcbf_dp_bucketed_sync_and_global_logs():
     """Bucketed (overlapped) all-reduce keeps ranks in lockstep AND the
     logged update scalars are all-reduced so every rank logs the same
     global-batch values."""
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = _free_port()
-    ps = [ctx.Process(target=_worker_gcbf_bucketed_logs,
-                      args=(r, 2, q, port))
-          for r in range(2)]
-    for p in ps:
-        p.start()
-    results = {}
-    for _ in range(2):
-        rank, w, scalars = q.get(timeout=600)
-        results[rank] = (w, scalars)
-    for p in ps:
-        p.join(timeout=600)
-        assert p.exitcode == 0
+    results = _spawn_collect(_worker_gcbf_bucketed_logs)
     assert torch.allclose(results[0][0], results[1][0], atol=1e-6)
     s0, s1 = results[0][1], results[1][1]
     assert len(s0) == len(s1) > 0

@@ -73,6 +73,25 @@ def _worker_body(rank, world, q, port):
 def test_two_rank_dp_on_one_gpu():
     if not torch.cuda.is_available():
         pytest.skip("GPU required")
+    last = None
+    for attempt in range(2):
+        # two processes sharing one GPU occasionally die in ROCm context
+        # setup (ConnectionReset on the result queue, no Python
+        # traceback); retry once — a real logic failure reproduces
+        try:
+            results = _launch_pair()
+            break
+        except (ConnectionResetError, EOFError, OSError) as e:
+            last = e
+    else:
+        raise last
+    # averaged grads from identical broadcast weights keep ranks in
+    # lockstep (bf16 numerics are deterministic per rank pair here)
+    assert torch.allclose(results[0], results[1], atol=1e-5), \
+        (results[0] - results[1]).abs().max()
+
+
+def _launch_pair():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = _free_port()
@@ -80,15 +99,17 @@ def test_two_rank_dp_on_one_gpu():
           for r in range(2)]
     for p in ps:
         p.start()
-    results = {}
-    for _ in range(2):
-        rank, w = q.get(timeout=600)
-        assert not (isinstance(w, str) and w.startswith("ERROR:")), w
-        results[rank] = w
-    for p in ps:
-        p.join(timeout=600)
-        assert p.exitcode == 0
-    # averaged grads from identical broadcast weights keep ranks in
-    # lockstep (bf16 numerics are deterministic per rank pair here)
-    assert torch.allclose(results[0], results[1], atol=1e-5), \
-        (results[0] - results[1]).abs().max()
+    try:
+        results = {}
+        for _ in range(2):
+            rank, w = q.get(timeout=600)
+            assert not (isinstance(w, str) and w.startswith("ERROR:")), w
+            results[rank] = w
+        for p in ps:
+            p.join(timeout=600)
+            assert p.exitcode == 0
+        return results
+    finally:
+        for p in ps:
+            if p.is_alive():
+                p.terminate()

