@@ -74,11 +74,14 @@ def _self_launch_torchrun(args):
 def run(args):
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
         _self_launch_torchrun(args)
-    rank, world_size, local_rank = init_distributed()
-    if world_size != args.gpus:
-        print(f"# FATAL: WORLD_SIZE={world_size} but --gpus {args.gpus}; "
+    env_world = int(os.environ.get("WORLD_SIZE", "1"))
+    if env_world != args.gpus:
+        # checked BEFORE init_process_group (which would block on the
+        # rendezvous): never report numbers under a mislabeled n_gpus
+        print(f"# FATAL: WORLD_SIZE={env_world} but --gpus {args.gpus}; "
               f"refusing to report mislabeled numbers", flush=True)
         raise SystemExit(2)
+    rank, world_size, local_rank = init_distributed()
     set_seed(args.seed + rank)
 
     use_cuda = torch.cuda.is_available()

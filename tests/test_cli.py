@@ -81,6 +81,22 @@ def test_bench_contract():
     assert "ms_per_step" in j and "config" in j
 
 
+def test_bench_refuses_mislabeled_world():
+    """Driver contract guard: WORLD_SIZE inconsistent with --gpus must
+    exit non-zero instead of printing mislabeled numbers."""
+    import os
+    env = dict(os.environ)
+    env.update({"WORLD_SIZE": "2", "RANK": "0", "LOCAL_RANK": "0",
+                "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29400"})
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "1", "--steps", "4",
+         "--warmup", "1", "--batch-size", "8", "-n", "4"],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 2, (r.returncode, r.stdout, r.stderr[-500:])
+    assert "refusing" in r.stdout
+
+
 def test_nominal_test_cli(tmp_path):
     r = _run([sys.executable, "test.py", "--env", "SimpleCar", "-n", "3",
               "--epi", "1", "--no-video", "--cpu"], timeout=1800)
