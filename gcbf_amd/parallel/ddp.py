@@ -141,7 +141,11 @@ class BucketedGradSynchronizer:
 
     Usage: install() once after building the modules; call the object
     (the ``grad_sync`` hook) between backward and the optimizer step to
-    flush stragglers and wait for completion.
+    flush stragglers and wait for completion.  The per-bucket pending
+    counters assume every backward is followed by exactly one __call__;
+    an exception between backward and the hook aborts the whole update
+    (the trainer re-raises), so partially-launched buckets are never
+    consumed.
     """
 
     BUCKET_BYTES = 25 * 1024 * 1024
