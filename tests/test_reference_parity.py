@@ -250,3 +250,130 @@ def test_dubins_masks_and_uref_parity(ref):
         m_ref = getattr(ref_env, name)(d_ref)
         m_my = getattr(my_env, name)(d_my)
         assert torch.equal(m_ref.bool(), m_my.bool()), name
+
+
+def test_macbf_nets_match_reference(ref):
+    """CBFNet (per-edge h) and MACBFController (max-aggregation) vs the
+    executing reference on identical weights and graphs."""
+    import importlib
+    ref_macbf = importlib.import_module("gcbf.algo.macbf")
+    ref_mc = importlib.import_module("gcbf.controller.macbf_controller")
+    from gcbf_amd.algo.macbf import CBFNet as MyCBFNet
+    from gcbf_amd.controller import MACBFController as MyMC
+    from gcbf_amd.graph import GraphBatch
+
+    torch.manual_seed(2)
+    r_cbf = ref_macbf.CBFNet(num_agents=8, node_dim=4, edge_dim=4)
+    m_cbf = MyCBFNet(num_agents=8, node_dim=4, edge_dim=4)
+    m_cbf.load_state_dict(r_cbf.state_dict(), strict=True)
+    r_ctrl = ref_mc.MACBFController(num_agents=8, node_dim=4, edge_dim=4,
+                                    phi_dim=128, action_dim=2)
+    m_ctrl = MyMC(num_agents=8, node_dim=4, edge_dim=4, phi_dim=128,
+                  action_dim=2)
+    m_ctrl.load_state_dict(r_ctrl.state_dict(), strict=True)
+    for m in (r_cbf, m_cbf, r_ctrl, m_ctrl):
+        m.eval()
+
+    x, states, ei, ea = _random_graph(n=8, node_dim=4, edge_dim=4, seed=5)
+    u_ref = torch.randn(8, 2)
+    Data = sys.modules["torch_geometric.data"].Data
+    d_ref = Data(x=x, pos=states[:, :2], states=states, edge_index=ei,
+                 edge_attr=ea, u_ref=u_ref)
+    d_my = GraphBatch(x=x, pos=states[:, :2], states=states, edge_index=ei,
+                      edge_attr=ea, u_ref=u_ref)
+    with torch.no_grad():
+        h_ref = r_cbf(d_ref)
+        h_my = m_cbf(d_my)
+        a_ref = r_ctrl(d_ref)
+        a_my = m_ctrl(d_my)
+    assert torch.allclose(h_ref, h_my, atol=1e-5)
+    assert torch.allclose(a_ref, a_my, atol=1e-5)
+
+
+def test_simple_drone_parity(ref):
+    """SimpleDrone graph build / u_ref / masks vs the executing reference
+    on an injected state (agents + obstacle nodes)."""
+    import importlib
+    ref_sd = importlib.import_module("gcbf.env.simple_drone")
+    from gcbf_amd.env import make_env
+    from gcbf_amd.graph import GraphBatch
+    torch.manual_seed(13)
+    dev = torch.device("cpu")
+    n = 6
+    my_env = make_env("SimpleDrone", n, dev)
+    ref_env = ref_sd.SimpleDrone(n, dev)
+    my_env.train()
+    ref_env.train()
+
+    side = my_env.default_params["area_size"]
+    ag = torch.cat([torch.rand(n, 3) * side, 0.2 * torch.randn(n, 3)],
+                   dim=1)
+    # the reference spawns num_agents obstacles regardless of num_obs
+    obs = torch.zeros(n, 6)
+    obs[:, :3] = torch.rand(n, 3) * side
+    states = torch.cat([ag, obs], dim=0)
+    goal = torch.cat([torch.rand(n, 3) * side, torch.zeros(n, 3)], dim=1)
+    agent_mask = torch.zeros(2 * n, dtype=torch.bool)
+    agent_mask[:n] = True
+    x = torch.cat([torch.zeros(n, 4), torch.ones(n, 4)], dim=0)
+
+    Data = sys.modules["torch_geometric.data"].Data
+    ref_env._goal = goal.clone()
+    my_env._goal = goal.clone()
+    ref_env._obs = obs.clone()
+    my_env._obs = obs.clone()
+    d_ref = ref_env.add_communication_links(
+        Data(x=x, pos=states[:, :3], states=states.clone(),
+             agent_mask=agent_mask))
+    gb = GraphBatch(x=x, pos=states[:, :3], states=states.clone(),
+                    agent_mask=agent_mask)
+    gb.agents_first_n = n
+    d_my = my_env.add_communication_links(gb)
+
+    e_ref = set(map(tuple, d_ref.edge_index.t().tolist()))
+    e_my = set(map(tuple, d_my.edge_index.t().tolist()))
+    assert e_ref == e_my
+
+    u_ref = ref_env.u_ref(d_ref)
+    u_my = my_env.u_ref(d_my)
+    assert torch.allclose(u_ref, u_my, atol=1e-5), \
+        (u_ref - u_my).abs().max()
+    for name in ("safe_mask", "unsafe_mask", "collision_mask"):
+        m_ref = getattr(ref_env, name)(d_ref)
+        m_my = getattr(my_env, name)(d_my)
+        assert torch.equal(m_ref.bool(), m_my.bool()), name
+
+
+def test_buffer_sampling_matches_reference(ref):
+    """Balanced segment sampling draws the SAME indices as the reference
+    under the same RNG state (np.random + random.choices call order)."""
+    import importlib
+    import random
+
+    import numpy as np
+    ref_buffer = importlib.import_module("gcbf.algo.buffer")
+    from gcbf_amd.algo.buffer import Buffer as MyBuffer
+
+    class _Tag:  # minimal graph stand-in (both buffers store objects)
+        def __init__(self, i):
+            self.i = i
+        ring_id = None
+
+    def fill(buf):
+        for i in range(200):
+            buf.append(_Tag(i), is_safe=(i % 3 != 0))
+
+    rb = ref_buffer.Buffer()
+    mb = MyBuffer()
+    fill(rb)
+    fill(mb)
+    for balanced in (False, True):
+        np.random.seed(42)
+        random.seed(43)
+        ref_out = rb.sample(20, 3, balanced) if balanced else \
+            rb.sample(20, 3)
+        np.random.seed(42)
+        random.seed(43)
+        my_out = mb.sample(20, 3, balanced) if balanced else \
+            mb.sample(20, 3)
+        assert [g.i for g in ref_out] == [g.i for g in my_out], balanced
