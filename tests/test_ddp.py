@@ -10,6 +10,14 @@ import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
+
+def _np(t):
+    """Queue-safe copy: torch.multiprocessing reduces tensors through
+    shared-memory FDs, and a worker exiting right after q.put races the
+    parent's mapping (ConnectionReset, seen under load) — numpy arrays
+    pickle inline."""
+    return t.detach().cpu().numpy().copy()
+
 from gcbf_amd.parallel import GradSynchronizer, broadcast_modules
 
 
@@ -74,16 +82,17 @@ def _worker_grad_sync(rank, world, q, port):
     local_grad = m.weight.grad.clone()
     sync = GradSynchronizer([m])
     sync()
-    q.put((rank, local_grad, m.weight.grad.clone()))
+    q.put((rank, _np(local_grad), _np(m.weight.grad)))
     dist.barrier()
     dist.destroy_process_group()
 
 
 def test_grad_synchronizer_averages():
     results = _spawn_collect(_worker_grad_sync, timeout=120)
+    import numpy as np
     mean = (results[0][0] + results[1][0]) / 2
-    assert torch.allclose(results[0][1], mean, atol=1e-6)
-    assert torch.allclose(results[1][1], mean, atol=1e-6)
+    assert np.allclose(results[0][1], mean, atol=1e-6)
+    assert np.allclose(results[1][1], mean, atol=1e-6)
 
 
 def _worker_gcbf_dp(rank, world, q, port):
@@ -109,16 +118,16 @@ def _worker_gcbf_dp(rank, world, q, port):
         if algo.is_update(step):
             algo.update(step, None)
     # after synced updates from identical init, weights must match
-    w = algo.actor.feat_2_action.net[0].weight.detach().clone()
-    q.put((rank, w))
+    q.put((rank, _np(algo.actor.feat_2_action.net[0].weight)))
     dist.barrier()
     dist.destroy_process_group()
 
 
 def test_gcbf_dp_two_ranks_stay_in_sync():
+    import numpy as np
     results = _spawn_collect(_worker_gcbf_dp)
     # identical optimizer trajectories (same averaged grads every step)
-    assert torch.allclose(results[0][0], results[1][0], atol=1e-6)
+    assert np.allclose(results[0][0], results[1][0], atol=1e-6)
 
 
 def _worker_bucketed(rank, world, q, port):
@@ -134,20 +143,21 @@ def _worker_bucketed(rank, world, q, port):
     x = torch.randn(4, 16)
     loss = m(x).pow(2).sum()
     loss.backward()
-    local = [p.grad.clone() for p in m.parameters()]
+    local = [_np(p.grad) for p in m.parameters()]
     sync()
-    q.put((rank, local, [p.grad.clone() for p in m.parameters()]))
+    q.put((rank, local, [_np(p.grad) for p in m.parameters()]))
     dist.barrier()
     dist.destroy_process_group()
 
 
 def test_bucketed_grad_synchronizer_averages():
+    import numpy as np
     results = _spawn_collect(_worker_bucketed, timeout=120)
     for g0, g1, s0, s1 in zip(results[0][0], results[1][0],
                               results[0][1], results[1][1]):
         mean = (g0 + g1) / 2
-        assert torch.allclose(s0, mean, atol=1e-6)
-        assert torch.allclose(s1, mean, atol=1e-6)
+        assert np.allclose(s0, mean, atol=1e-6)
+        assert np.allclose(s1, mean, atol=1e-6)
 
 
 class _CollectWriter:
@@ -182,8 +192,8 @@ def _worker_gcbf_bucketed_logs(rank, world, q, port):
             data = env.reset()
         if algo.is_update(step):
             algo.update(step, writer)
-    w = algo.actor.feat_2_action.net[0].weight.detach().clone()
-    q.put((rank, w, writer.scalars))
+    q.put((rank, _np(algo.actor.feat_2_action.net[0].weight),
+           writer.scalars))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -192,8 +202,9 @@ def test_gcbf_dp_bucketed_sync_and_global_logs():
     """Bucketed (overlapped) all-reduce keeps ranks in lockstep AND the
     logged update scalars are all-reduced so every rank logs the same
     global-batch values."""
+    import numpy as np
     results = _spawn_collect(_worker_gcbf_bucketed_logs)
-    assert torch.allclose(results[0][0], results[1][0], atol=1e-6)
+    assert np.allclose(results[0][0], results[1][0], atol=1e-6)
     s0, s1 = results[0][1], results[1][1]
     assert len(s0) == len(s1) > 0
     for (n0, v0, t0), (n1, v1, t1) in zip(s0, s1):

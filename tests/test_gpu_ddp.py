@@ -65,7 +65,9 @@ def _worker_body(rank, world, q, port):
         if algo.is_update(step):
             algo.update(step, None)
     w = algo.actor.feat_2_action.net[0].weight.detach().float().cpu()
-    q.put((rank, w))
+    # numpy copy: tensor queue reduction uses shared-memory FDs and races
+    # worker exit (ConnectionReset)
+    q.put((rank, w.numpy().copy()))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -87,8 +89,9 @@ def test_two_rank_dp_on_one_gpu():
         raise last
     # averaged grads from identical broadcast weights keep ranks in
     # lockstep (bf16 numerics are deterministic per rank pair here)
-    assert torch.allclose(results[0], results[1], atol=1e-5), \
-        (results[0] - results[1]).abs().max()
+    import numpy as np
+    assert np.allclose(results[0], results[1], atol=1e-5), \
+        np.abs(results[0] - results[1]).max()
 
 
 def _launch_pair():
