@@ -240,6 +240,12 @@ class GCBF(Algorithm):
                 prev = self.memory.sample(
                     self.batch_size // 5 - self.batch_size // 10, seg_len, True)
                 graph_list = curr + prev
+            if not graph_list:
+                # degenerate batch_size (< 10): the reference's balanced
+                # quotas (n//2 draws) collapse to zero and it crashes at
+                # Batch.from_data_list([]); sample unbalanced instead
+                graph_list = self.buffer.sample(
+                    max(1, self.batch_size // 5), seg_len)
             if prof:
                 t1 = _tick(); prof["sample"] += t1 - t0; t0 = t1
 
