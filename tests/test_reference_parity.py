@@ -377,3 +377,29 @@ def test_buffer_sampling_matches_reference(ref):
         my_out = mb.sample(20, 3, balanced) if balanced else \
             mb.sample(20, 3)
         assert [g.i for g in ref_out] == [g.i for g in my_out], balanced
+
+
+def test_attention_matches_reference(ref):
+    """CBFGNNLayer.attention (the plot_cbf path: per-edge softmax gate)
+    vs the executing reference on shared weights."""
+    from gcbf_amd.algo.gcbf import CBFGNN as MyCBFGNN
+    from gcbf_amd.graph import GraphBatch
+    torch.manual_seed(21)
+    ref_net = ref["gcbf"].CBFGNN(num_agents=8, node_dim=4, edge_dim=5,
+                                 phi_dim=256)
+    mine = MyCBFGNN(num_agents=8, node_dim=4, edge_dim=5, phi_dim=256)
+    mine.load_state_dict(ref_net.state_dict(), strict=True)
+    ref_net.eval()
+    mine.eval()
+    x, states, ei, ea = _random_graph(seed=23)
+    Data = sys.modules["torch_geometric.data"].Data
+    d_ref = Data(x=x, pos=states[:, :2], states=states, edge_index=ei,
+                 edge_attr=ea)
+    d_my = GraphBatch(x=x, pos=states[:, :2], states=states, edge_index=ei,
+                      edge_attr=ea)
+    with torch.no_grad():
+        a_ref = ref_net.attention(d_ref)
+        a_my = mine.attention(d_my)
+    assert a_ref.shape == a_my.shape
+    assert torch.allclose(a_ref, a_my, atol=1e-5), \
+        (a_ref - a_my).abs().max()
