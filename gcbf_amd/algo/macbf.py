@@ -138,7 +138,14 @@ class MACBF(GCBF):
                 loss_action.detach(), acc_unsafe.detach(), acc_safe.detach(),
                 acc_h_dot.detach()]))
 
-        log_vals = torch.stack(logs).cpu()
+        # mean-reduce across DP ranks so rank-0 curves reflect the global
+        # batch (same semantics as GCBF._update_tail)
+        log_stack = torch.stack(logs).detach()
+        import torch.distributed as dist
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(log_stack, op=dist.ReduceOp.SUM)
+            log_stack /= dist.get_world_size()
+        log_vals = log_stack.cpu()
         if writer is not None:
             names = ("loss/unsafe", "loss/safe", "loss/derivative",
                      "loss/action", "acc/unsafe", "acc/safe", "acc/derivative")
